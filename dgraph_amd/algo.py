@@ -1,0 +1,307 @@
+"""Host-side mirror of dgraph's `algo` package over the uidalgo C-ABI.
+
+Function-for-function surface of /root/reference/algo/uidlist.go (same names'
+semantics, same argument meaning, status-code errors), plus the batched
+device-resident engine the reference's per-key fan-out maps onto
+(worker/task.go:834-971 -> one grid).
+
+UID lists are sorted uint64.  torch has no uint64 CUDA dtype, so device
+buffers are int64 tensors REINTERPRETED as u64 by the kernels; numpy uint64
+arrays cross the host boundary bit-for-bit via .view(int64).
+
+No CPU fallback: every op goes through libuidalgo.so and raises without it.
+"""
+import ctypes as C
+
+import numpy as np
+
+from dgraph_amd._lib import (UaDPack, UaDPair, check, lib, _u64p)
+
+_u64 = C.c_uint64
+
+
+def _np_u64(x):
+    a = np.ascontiguousarray(np.asarray(x, dtype=np.uint64))
+    return a
+
+
+_KEEP = np.empty(1, dtype=np.uint64)
+
+
+def _hptr(a):
+    if a.size == 0:
+        return _KEEP.ctypes.data_as(_u64p)
+    return a.ctypes.data_as(_u64p)
+
+
+class Engine:
+    """One GPU + one HIP stream (wraps ua_ctx)."""
+
+    def __init__(self, device=0):
+        self._ctx = C.c_void_p()
+        rc = lib().ua_ctx_create(C.byref(self._ctx), device)
+        if rc != 0:
+            raise RuntimeError(
+                f"uidalgo: cannot create GPU context on device {device}: "
+                f"{lib().ua_strerror(rc).decode()} — the HIP engine is "
+                "mandatory, there is no CPU fallback.")
+        self.device = device
+
+    def close(self):
+        if self._ctx:
+            lib().ua_ctx_destroy(self._ctx)
+            self._ctx = C.c_void_p()
+
+    def __del__(self):
+        try:
+            self.close()
+        except Exception:
+            pass
+
+    # ---- stats (dominant-kernel HIP-event time) ----
+    def stats_reset(self):
+        check(lib().ua_stats_reset(self._ctx))
+
+    def stats(self):
+        n = _u64()
+        ms = C.c_double()
+        b = _u64()
+        check(lib().ua_stats_get(self._ctx, C.byref(n), C.byref(ms), C.byref(b)))
+        return {"launches": n.value, "kernel_ms": ms.value, "bytes_algorithmic": b.value}
+
+    # ---- batched device-resident ops (torch CUDA int64 tensors) ----
+    def _run_pairs(self, fn, us, vs, outs):
+        np_ = len(us)
+        pairs = (UaDPair * np_)()
+        for i in range(np_):
+            pairs[i].u = us[i].data_ptr()
+            pairs[i].n = us[i].numel()
+            pairs[i].v = vs[i].data_ptr()
+            pairs[i].m = vs[i].numel()
+            pairs[i].out = outs[i].data_ptr()
+        lens = (_u64 * np_)()
+        check(fn(self._ctx, pairs, np_, lens))
+        return [int(lens[i]) for i in range(np_)]
+
+    def intersect_pairs(self, us, vs, outs=None):
+        """Batched algo.IntersectWith.  us/vs: lists of CUDA int64 tensors
+        (u64 bits).  Returns (outs, lens); outs[i][:lens[i]] is the result."""
+        import torch
+        if outs is None:
+            outs = [torch.empty(min(u.numel(), v.numel()), dtype=torch.int64,
+                                device=u.device) for u, v in zip(us, vs)]
+        lens = self._run_pairs(lib().ua_intersect_batch_dev, us, vs, outs)
+        return outs, lens
+
+    def merge_pairs(self, us, vs, outs=None):
+        import torch
+        if outs is None:
+            outs = [torch.empty(u.numel() + v.numel(), dtype=torch.int64,
+                                device=u.device) for u, v in zip(us, vs)]
+        lens = self._run_pairs(lib().ua_merge_batch_dev, us, vs, outs)
+        return outs, lens
+
+    def difference_pairs(self, us, vs, outs=None):
+        import torch
+        if outs is None:
+            outs = [torch.empty(max(u.numel(), 1), dtype=torch.int64, device=u.device)
+                    for u in us]
+        lens = self._run_pairs(lib().ua_difference_batch_dev, us, vs, outs)
+        return outs, lens
+
+    def intersect_sorted(self, lists):
+        """algo.IntersectSorted (uidlist.go:297): k-way fold, smallest first."""
+        import torch
+        k = len(lists)
+        if k == 0:
+            return torch.empty(0, dtype=torch.int64)
+        ptrs = (C.c_void_p * k)(*[int(t.data_ptr()) for t in lists])
+        lens = (_u64 * k)(*[t.numel() for t in lists])
+        cap = min(t.numel() for t in lists)
+        out = torch.empty(max(cap, 1), dtype=torch.int64, device=lists[0].device)
+        out_n = _u64()
+        check(lib().ua_intersect_k_dev(self._ctx, C.cast(ptrs, C.POINTER(C.c_void_p)),
+                                       lens, k, C.c_void_p(out.data_ptr()),
+                                       C.byref(out_n)))
+        return out[:out_n.value]
+
+    def merge_sorted(self, lists):
+        """algo.MergeSorted (uidlist.go:448): dedup'd k-way union."""
+        import torch
+        k = len(lists)
+        if k == 0:
+            return torch.empty(0, dtype=torch.int64)
+        ptrs = (C.c_void_p * k)(*[int(t.data_ptr()) for t in lists])
+        lens = (_u64 * k)(*[t.numel() for t in lists])
+        cap = sum(t.numel() for t in lists)
+        out = torch.empty(max(cap, 1), dtype=torch.int64, device=lists[0].device)
+        out_n = _u64()
+        check(lib().ua_merge_k_dev(self._ctx, C.cast(ptrs, C.POINTER(C.c_void_p)),
+                                   lens, k, C.c_void_p(out.data_ptr()), C.byref(out_n)))
+        return out[:out_n.value]
+
+    def index_of_batch(self, u, queries):
+        """Batched algo.IndexOf: u, queries CUDA int64; returns int64 tensor
+        of positions (-1 = absent)."""
+        import torch
+        out = torch.empty(queries.numel(), dtype=torch.int64, device=u.device)
+        check(lib().ua_index_of_batch_dev(
+            self._ctx, C.c_void_p(u.data_ptr()), u.numel(),
+            C.c_void_p(queries.data_ptr()), queries.numel(),
+            C.c_void_p(out.data_ptr())))
+        return out
+
+    # ---- packed (codec) path ----
+    def upload_pack(self, bases, num_uids, delta_offs, deltas_blob, block_size):
+        """Upload flat pack arrays (from encode_flat) -> DPack on this GPU."""
+        import torch
+        dev = f"cuda:{self.device}"
+        t_bases = torch.from_numpy(_np_u64(bases).view(np.int64)).to(dev)
+        t_nums = torch.from_numpy(np.ascontiguousarray(num_uids, dtype=np.uint32)
+                                  .view(np.int32)).to(dev)
+        t_offs = torch.from_numpy(_np_u64(delta_offs).view(np.int64)).to(dev)
+        blob = np.ascontiguousarray(deltas_blob, dtype=np.uint8)
+        t_blob = torch.from_numpy(blob.view(np.int8)).to(dev) if blob.size else \
+            torch.zeros(1, dtype=torch.int8, device=dev)
+        total = int(np.asarray(num_uids, dtype=np.uint64).sum())
+        return DPack(t_bases, t_nums, t_offs, t_blob, int(block_size), total)
+
+    def intersect_packed(self, dpack, after, v, out=None):
+        """algo.IntersectCompressedWith (uidlist.go:33): fused decode+intersect."""
+        import torch
+        if out is None:
+            out = torch.empty(max(min(dpack.total_uids, v.numel()), 1),
+                              dtype=torch.int64, device=v.device)
+        out_n = _u64()
+        pk = dpack.struct()
+        check(lib().ua_intersect_packed_dev(
+            self._ctx, C.byref(pk), _u64(after), C.c_void_p(v.data_ptr()),
+            v.numel(), C.c_void_p(out.data_ptr()), C.byref(out_n)))
+        return out[:out_n.value]
+
+    def decode_pack(self, dpack, seek=0, out=None):
+        """codec.Decode(pack, seek) on the GPU (codec.go:444)."""
+        import torch
+        if out is None:
+            out = torch.empty(max(dpack.total_uids, 1), dtype=torch.int64,
+                              device=dpack.bases.device)
+        out_n = _u64()
+        pk = dpack.struct()
+        check(lib().ua_decode_dev(self._ctx, C.byref(pk), _u64(seek),
+                                  C.c_void_p(out.data_ptr()), C.byref(out_n)))
+        return out[:out_n.value]
+
+
+class DPack:
+    """Flattened pb.UidPack resident in HBM (engine-native layout)."""
+
+    def __init__(self, bases, num_uids, delta_offs, deltas, block_size, total_uids):
+        self.bases = bases
+        self.num_uids = num_uids
+        self.delta_offs = delta_offs
+        self.deltas = deltas
+        self.block_size = block_size
+        self.total_uids = total_uids
+
+    def struct(self):
+        pk = UaDPack()
+        pk.block_size = self.block_size
+        pk.n_blocks = self.bases.numel()
+        pk.bases = self.bases.data_ptr()
+        pk.num_uids = self.num_uids.data_ptr()
+        pk.delta_offs = self.delta_offs.data_ptr()
+        pk.deltas = self.deltas.data_ptr()
+        pk.total_uids = self.total_uids
+        return pk
+
+
+# ---- host-side codec (product encoder; codec.Encode restated in C++) ----
+
+def encode_flat(uids, block_size=256):
+    """codec.Encode(uids, blockSize) -> flat arrays
+    (bases u64[nb], num_uids u32[nb], delta_offs u64[nb+1], deltas u8[...])."""
+    uids = _np_u64(uids)
+    h = C.c_void_p()
+    check(lib().ua_encode(_hptr(uids), uids.size, block_size, C.byref(h)))
+    try:
+        view = lib().ua_owned_pack_view(h)
+        nb = _u64()
+        db = _u64()
+        tu = _u64()
+        check(lib().ua_pack_flat_sizes(view, C.byref(nb), C.byref(db), C.byref(tu)))
+        bases = np.empty(nb.value, dtype=np.uint64)
+        nums = np.empty(nb.value, dtype=np.uint32)
+        offs = np.empty(nb.value + 1, dtype=np.uint64)
+        blob = np.empty(max(db.value, 1), dtype=np.uint8)
+        check(lib().ua_pack_flatten(
+            view, _hptr(bases), nums.ctypes.data_as(C.POINTER(C.c_uint32)),
+            _hptr(offs), blob.ctypes.data_as(C.POINTER(C.c_uint8))))
+        return bases, nums, offs, blob[:db.value], tu.value
+    finally:
+        lib().ua_owned_pack_free(h)
+
+
+# ---- host-pointer convenience mirror (the cgo drop-in surface) ----
+# These run on the GPU via upload/compute/download; they exist so the cgo shim
+# maps 1:1 onto algo's signatures.
+
+def intersect_with(engine, u, v):
+    u, v = _np_u64(u), _np_u64(v)
+    out = np.empty(max(min(u.size, v.size), 1), dtype=np.uint64)
+    n = _u64()
+    check(lib().ua_intersect(engine._ctx, _hptr(u), u.size, _hptr(v), v.size,
+                             _hptr(out), C.byref(n)))
+    return out[:n.value]
+
+
+def difference(engine, u, v):
+    u, v = _np_u64(u), _np_u64(v)
+    out = np.empty(max(u.size, 1), dtype=np.uint64)
+    n = _u64()
+    check(lib().ua_difference(engine._ctx, _hptr(u), u.size, _hptr(v), v.size,
+                              _hptr(out), C.byref(n)))
+    return out[:n.value]
+
+
+def _host_lists(lists):
+    arrs = [_np_u64(x) for x in lists]
+    k = len(arrs)
+    ptrs = (_u64p * max(k, 1))(*[_hptr(a) for a in arrs])
+    lens = (_u64 * max(k, 1))(*[a.size for a in arrs])
+    return arrs, ptrs, lens, k
+
+
+def intersect_sorted(engine, lists):
+    arrs, ptrs, lens, k = _host_lists(lists)
+    cap = min((a.size for a in arrs), default=0)
+    out = np.empty(max(cap, 1), dtype=np.uint64)
+    n = _u64()
+    check(lib().ua_intersect_k(engine._ctx, ptrs, lens, k, _hptr(out), C.byref(n)))
+    return out[:n.value]
+
+
+def merge_sorted(engine, lists):
+    arrs, ptrs, lens, k = _host_lists(lists)
+    cap = sum(a.size for a in arrs)
+    out = np.empty(max(cap, 1), dtype=np.uint64)
+    n = _u64()
+    check(lib().ua_merge_k(engine._ctx, ptrs, lens, k, _hptr(out), C.byref(n)))
+    return out[:n.value]
+
+
+def index_of(u, uid):
+    """algo.IndexOf — host binary search, like the reference (uidlist.go:546)."""
+    u = _np_u64(u)
+    return int(lib().ua_index_of(_hptr(u), u.size, _u64(uid)))
+
+
+def apply_filter(u, mask):
+    """algo.ApplyFilter (uidlist.go:21): boolean-mask compaction.
+
+    The reference takes a Go closure; across a C ABI the filter arrives as a
+    precomputed mask (the callers evaluate per-uid predicates upstream)."""
+    import torch
+    if isinstance(u, torch.Tensor):
+        return u[mask]
+    u = _np_u64(u)
+    return u[np.asarray(mask, dtype=bool)]

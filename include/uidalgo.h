@@ -1,0 +1,174 @@
+/*
+ * uidalgo — C-ABI boundary of the MI355X-native posting-list set-algebra
+ * engine (libuidalgo.so).
+ *
+ * This is the drop-in surface for dgraph's `algo` package (reference
+ * interfaces each export replaces are cited below; see INTEGRATION.md for the
+ * cgo binding a dgraph maintainer would add).  All memory is caller-owned,
+ * errors are int status codes, a ua_ctx wraps one GPU + one HIP stream and is
+ * callable from any thread (calls on one ctx are serialized by its stream).
+ *
+ * Two levels:
+ *  - host-pointer convenience calls that mirror algo's signatures 1:1
+ *    (upload + compute + download; PCIe-inclusive), and
+ *  - device-resident batched calls (`*_dev`), the batched engine that a
+ *    worker/task.go:834-971 fan-out would feed: one grid per
+ *    SubGraph's thousands of per-key list ops.
+ *
+ * No torch types, no C++ types: plain pointers and sizes only.
+ */
+#ifndef UIDALGO_H
+#define UIDALGO_H
+
+#include <stddef.h>
+#include <stdint.h>
+
+#ifdef __cplusplus
+extern "C" {
+#endif
+
+/* ---- status codes ---- */
+enum {
+    UA_OK = 0,
+    UA_ERR_HIP = 1,      /* HIP runtime error (see ua_last_hip_error) */
+    UA_ERR_NOMEM = 2,    /* allocation failed */
+    UA_ERR_INVALID = 3,  /* bad argument (e.g. block with num_uids > 256) */
+    UA_ERR_NO_GPU = 4,   /* no HIP device visible */
+};
+
+const char *ua_strerror(int code);
+int ua_version(void);
+
+/* ---- context ---- */
+typedef struct ua_ctx ua_ctx;
+int ua_ctx_create(ua_ctx **out, int device);
+void ua_ctx_destroy(ua_ctx *ctx);
+
+/* ---- device memory (for callers without their own allocator) ---- */
+int ua_dev_alloc(ua_ctx *, uint64_t bytes, void **dptr);
+int ua_dev_free(ua_ctx *, void *dptr);
+int ua_h2d(ua_ctx *, void *dst_dev, const void *src_host, uint64_t bytes);
+int ua_d2h(ua_ctx *, void *dst_host, const void *src_dev, uint64_t bytes);
+int ua_sync(ua_ctx *);
+
+/* ---- engine stats (HIP-event timing of the dominant kernels, for the
+ * roofline leg of bench.py: kernel_ms is accumulated on the engine's own
+ * stream across launches since the last reset) ---- */
+int ua_stats_reset(ua_ctx *);
+int ua_stats_get(ua_ctx *, uint64_t *n_launches, double *kernel_ms,
+                 uint64_t *bytes_algorithmic);
+
+/* ---- pb.UidPack mirror (pb.proto:379-400) ---- */
+typedef struct {
+    uint64_t base;          /* pb.UidBlock.base */
+    uint32_t num_uids;      /* pb.UidBlock.num_uids (includes base) */
+    uint32_t deltas_len;
+    const uint8_t *deltas;  /* group-varint deltas */
+} ua_block;
+
+typedef struct {
+    uint32_t block_size;    /* pb.UidPack.block_size */
+    uint64_t n_blocks;
+    const ua_block *blocks;
+} ua_pack;
+
+/* Flattened pack in device memory — the engine's native layout. */
+typedef struct {
+    uint32_t block_size;
+    uint64_t n_blocks;
+    const uint64_t *bases;      /* device [n_blocks] */
+    const uint32_t *num_uids;   /* device [n_blocks] */
+    const uint64_t *delta_offs; /* device [n_blocks+1], offsets into deltas */
+    const uint8_t *deltas;      /* device blob */
+    uint64_t total_uids;        /* codec.ExactLen */
+} ua_dpack;
+
+/* ---- host-side codec (replaces codec.Encode, codec.go:393) ---- */
+typedef struct ua_owned_pack ua_owned_pack;
+int ua_encode(const uint64_t *uids, uint64_t n, uint32_t block_size,
+              ua_owned_pack **out);
+const ua_pack *ua_owned_pack_view(ua_owned_pack *);
+void ua_owned_pack_free(ua_owned_pack *);
+
+uint64_t ua_pack_exact_len(const ua_pack *);  /* codec.ExactLen  codec.go:427 */
+uint64_t ua_pack_approx_len(const ua_pack *); /* codec.ApproxLen codec.go:418 */
+
+/* Flatten a host pack into the engine layout (arrays caller-allocated:
+ * bases[n_blocks], num_uids[n_blocks], delta_offs[n_blocks+1],
+ * deltas_blob[total deltas bytes]).  Query sizes first. */
+int ua_pack_flat_sizes(const ua_pack *, uint64_t *n_blocks,
+                       uint64_t *deltas_bytes, uint64_t *total_uids);
+int ua_pack_flatten(const ua_pack *, uint64_t *bases, uint32_t *num_uids,
+                    uint64_t *delta_offs, uint8_t *deltas_blob);
+
+/* Decode a pack on the GPU (replaces codec.Decode, codec.go:444).
+ * out: device buffer, capacity total_uids. */
+int ua_decode_dev(ua_ctx *, const ua_dpack *, uint64_t seek_uid,
+                  uint64_t *out, uint64_t *out_n);
+
+/* ---- batched device-resident set algebra ----
+ * All pointers in ua_dpair are DEVICE pointers; the descriptor array itself
+ * and out_lens live on the host.  Inputs are sorted uint64 lists; results are
+ * bit-exact vs the reference on duplicate-free inputs (the parity domain
+ * pinned by uidlist_test.go:394,536-542 — see DESIGN.md §semantics). */
+typedef struct {
+    const uint64_t *u; /* device */
+    uint64_t n;
+    const uint64_t *v; /* device */
+    uint64_t m;
+    uint64_t *out;     /* device; capacity: intersect >= min(n,m),
+                        *                   merge >= n+m, difference >= n */
+} ua_dpair;
+
+/* batched algo.IntersectWith (uidlist.go:142) */
+int ua_intersect_batch_dev(ua_ctx *, const ua_dpair *pairs, int n_pairs,
+                           uint64_t *out_lens);
+/* batched pairwise algo.MergeSorted semantics (dedup union, uidlist.go:448) */
+int ua_merge_batch_dev(ua_ctx *, const ua_dpair *pairs, int n_pairs,
+                       uint64_t *out_lens);
+/* batched algo.Difference (uidlist.go:332) */
+int ua_difference_batch_dev(ua_ctx *, const ua_dpair *pairs, int n_pairs,
+                            uint64_t *out_lens);
+
+/* k-way fold algo.IntersectSorted (uidlist.go:297): lists/lens host arrays of
+ * device pointers; out device, capacity min(lens). */
+int ua_intersect_k_dev(ua_ctx *, const uint64_t *const *lists,
+                       const uint64_t *lens, int k, uint64_t *out,
+                       uint64_t *out_n);
+/* k-way algo.MergeSorted (uidlist.go:448): pairwise tree on device.
+ * out device, capacity sum(lens). */
+int ua_merge_k_dev(ua_ctx *, const uint64_t *const *lists,
+                   const uint64_t *lens, int k, uint64_t *out,
+                   uint64_t *out_n);
+
+/* batched algo.IndexOf (uidlist.go:546): queries/out device. */
+int ua_index_of_batch_dev(ua_ctx *, const uint64_t *u, uint64_t n,
+                          const uint64_t *queries, uint64_t nq, int64_t *out);
+
+/* fused decode+intersect: algo.IntersectCompressedWith (uidlist.go:33) over a
+ * device pack; v device; out device, capacity min(total_uids, m). */
+int ua_intersect_packed_dev(ua_ctx *, const ua_dpack *, uint64_t after_uid,
+                            const uint64_t *v, uint64_t m, uint64_t *out,
+                            uint64_t *out_n);
+
+/* ---- host-pointer convenience (upload+compute+download; mirrors the algo
+ * package signatures 1:1 for the cgo shim) ---- */
+int ua_intersect(ua_ctx *, const uint64_t *u, uint64_t n, const uint64_t *v,
+                 uint64_t m, uint64_t *out, uint64_t *out_n);  /* IntersectWith */
+int ua_intersect_k(ua_ctx *, const uint64_t *const *lists, const uint64_t *lens,
+                   int k, uint64_t *out, uint64_t *out_n);     /* IntersectSorted */
+int ua_merge_k(ua_ctx *, const uint64_t *const *lists, const uint64_t *lens,
+               int k, uint64_t *out, uint64_t *out_n);         /* MergeSorted */
+int ua_difference(ua_ctx *, const uint64_t *u, uint64_t n, const uint64_t *v,
+                  uint64_t m, uint64_t *out, uint64_t *out_n); /* Difference */
+int64_t ua_index_of(const uint64_t *u, uint64_t n, uint64_t uid); /* IndexOf:
+                  * one log(n) probe — host binary search like the reference;
+                  * the batched GPU form is ua_index_of_batch_dev */
+int ua_intersect_packed(ua_ctx *, const ua_pack *, uint64_t after_uid,
+                        const uint64_t *v, uint64_t m, uint64_t *out,
+                        uint64_t *out_n);            /* IntersectCompressedWith */
+
+#ifdef __cplusplus
+}
+#endif
+#endif /* UIDALGO_H */

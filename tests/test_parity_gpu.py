@@ -1,0 +1,308 @@
+"""GPU parity: the HIP engine vs the CPU oracle (bit-exact restatement of the
+Go reference) on the same seeded inputs.  The bar is BIT-EXACT equality on
+duplicate-free sorted inputs — the parity domain the reference itself pins
+(uidlist_test.go:394, :536-542).
+
+Everything here goes through the C-ABI of libuidalgo.so (the product path);
+the oracle appears only as the checker.
+"""
+import numpy as np
+import pytest
+import torch
+
+from dgraph_amd import algo, synth
+from oracle import bind as orc
+
+pytestmark = pytest.mark.gpu
+
+SEED = synth.SEED
+
+
+@pytest.fixture(scope="module")
+def eng():
+    e = algo.Engine(0)
+    yield e
+    e.close()
+
+
+def to_dev(a):
+    a = np.ascontiguousarray(a, dtype=np.uint64)
+    if a.size == 0:
+        return torch.empty(0, dtype=torch.int64, device="cuda:0")
+    return torch.from_numpy(a.view(np.int64)).to("cuda:0")
+
+
+def to_np(t):
+    return t.cpu().numpy().view(np.uint64)
+
+
+# ---------- golden duplicate-free cases through the host-convenience API ----------
+
+GOLDEN_INTERSECT = [
+    ([1, 2, 3], [], []),
+    ([1, 2, 3], [1, 2, 3, 4, 5], [1, 2, 3]),
+    ([1, 2, 3], [2], [2]),
+    ([1, 2, 3], [0, 5], []),
+    ([1, 2, 3], [3, 5], [3]),
+]
+
+
+@pytest.mark.parametrize("u,v,want", GOLDEN_INTERSECT)
+def test_golden_intersect_host_api(eng, u, v, want):
+    got = algo.intersect_with(eng, u, v)
+    assert got.tolist() == want
+
+
+GOLDEN_DIFF = [
+    ([1, 2, 3], [1], [2, 3]),
+    ([1, 2, 3], [2], [1, 3]),
+    ([1, 2, 3], [3], [1, 2]),
+    ([1, 2, 3], [], [1, 2, 3]),
+    ([], [1, 2], []),
+    ([1, 2, 3], [2, 3, 4, 5], [1]),
+    ([10, 12, 13], [2, 3, 4, 13], [10, 12]),
+]
+
+
+@pytest.mark.parametrize("u,v,want", GOLDEN_DIFF)
+def test_golden_difference_host_api(eng, u, v, want):
+    got = algo.difference(eng, u, v)
+    assert got.tolist() == want
+
+
+GOLDEN_MERGE = [
+    ([[55]], [55]),
+    ([[1, 3, 6, 8, 10], [2, 4, 5, 7, 15]], [1, 2, 3, 4, 5, 6, 7, 8, 10, 15]),
+    ([[1, 3, 6, 8, 10], []], [1, 3, 6, 8, 10]),
+    ([[], [1, 3, 6, 8, 10]], [1, 3, 6, 8, 10]),
+    ([[], []], []),
+    ([[5, 6, 7], [3, 4], [1, 2], []], [1, 2, 3, 4, 5, 6, 7]),
+    ([], []),
+]
+
+
+@pytest.mark.parametrize("lists,want", GOLDEN_MERGE)
+def test_golden_merge_host_api(eng, lists, want):
+    got = algo.merge_sorted(eng, lists)
+    assert got.tolist() == want
+
+
+GOLDEN_INTERSECT_K = [
+    ([[1, 2, 3], [2, 3, 4, 5]], [2, 3]),
+    ([[1, 2, 3]], [1, 2, 3]),
+    ([], []),
+    ([[100, 101]], [100, 101]),
+    ([[1, 2, 3], [2, 3, 4, 5], [4, 5, 6]], []),
+    ([[10, 12, 13], [2, 3, 4, 13], [4, 5, 6]], []),
+]
+
+
+@pytest.mark.parametrize("lists,want", GOLDEN_INTERSECT_K)
+def test_golden_intersect_k_host_api(eng, lists, want):
+    got = algo.intersect_sorted(eng, lists)
+    assert got.tolist() == want
+
+
+# ---------- randomized parity vs oracle, device-resident batched path ----------
+
+SIZES = [
+    (0, 0), (1, 1), (1, 1000), (10, 1_000_000),  # ratio 1e5 (the 766ns/op shape)
+    (1000, 1000), (4096, 100_000), (100_000, 100_000),
+    (1_000_000, 1_000_000),
+]
+
+
+@pytest.mark.parametrize("n,m", SIZES)
+def test_intersect_pair_vs_oracle(eng, n, m):
+    rng = np.random.default_rng(SEED + n * 7 + m)
+    limit = max((n + m) * 3, 100)
+    u = synth.gen_sorted_unique(rng, n, limit)
+    v = synth.gen_sorted_unique(rng, m, limit)
+    outs, lens = eng.intersect_pairs([to_dev(u)], [to_dev(v)])
+    got = to_np(outs[0][:lens[0]])
+    want = orc.intersect_with(u, v)
+    assert got.tolist() == want.tolist()
+
+
+@pytest.mark.parametrize("n,m", SIZES)
+def test_difference_pair_vs_oracle(eng, n, m):
+    rng = np.random.default_rng(SEED + n * 13 + m)
+    limit = max((n + m) * 3, 100)
+    u = synth.gen_sorted_unique(rng, n, limit)
+    v = synth.gen_sorted_unique(rng, m, limit)
+    outs, lens = eng.difference_pairs([to_dev(u)], [to_dev(v)])
+    got = to_np(outs[0][:lens[0]])
+    assert got.tolist() == orc.difference(u, v).tolist()
+
+
+@pytest.mark.parametrize("n,m", SIZES)
+def test_merge_pair_vs_oracle(eng, n, m):
+    rng = np.random.default_rng(SEED + n * 17 + m)
+    limit = max((n + m) * 3, 100)
+    u = synth.gen_sorted_unique(rng, n, limit)
+    v = synth.gen_sorted_unique(rng, m, limit)
+    outs, lens = eng.merge_pairs([to_dev(u)], [to_dev(v)])
+    got = to_np(outs[0][:lens[0]])
+    assert got.tolist() == orc.merge_sorted([u, v]).tolist()
+
+
+def test_cfg2_planted_overlap(eng):
+    """cfg 2: 1M x 1M, 1% planted overlap — exact output known by construction."""
+    rng = np.random.default_rng(SEED)
+    u, v, common = synth.gen_pair(rng, 1_000_000, 1_000_000, 10_000, 100_000_000)
+    outs, lens = eng.intersect_pairs([to_dev(u)], [to_dev(v)])
+    got = to_np(outs[0][:lens[0]])
+    assert lens[0] == common.size
+    assert np.array_equal(got, common)
+    # and vs oracle
+    assert np.array_equal(got, orc.intersect_with(u, v))
+
+
+def test_batched_zipf_vs_oracle(eng):
+    """cfg 3 (scaled): 96 Zipf-sized pairs in ONE grid, each bit-exact."""
+    rng = np.random.default_rng(SEED + 3)
+    sizes = synth.zipf_sizes(rng, 96, lo=100, hi=200_000)
+    us, vs, wants = [], [], []
+    for i, sz in enumerate(sizes):
+        n = int(sz)
+        m = int(max(100, n // (10 ** int(rng.integers(0, 3)))))
+        u = synth.gen_sorted_unique(rng, n, 3 * (n + m))
+        v = synth.gen_sorted_unique(rng, m, 3 * (n + m))
+        us.append(u)
+        vs.append(v)
+    d_us = [to_dev(x) for x in us]
+    d_vs = [to_dev(x) for x in vs]
+    outs, lens = eng.intersect_pairs(d_us, d_vs)
+    wants = orc.intersect_batch_cpu(us, vs)
+    for i in range(len(us)):
+        assert to_np(outs[i][:lens[i]]).tolist() == wants[i].tolist(), f"pair {i}"
+    # merge + difference over the same batch
+    m_outs, m_lens = eng.merge_pairs(d_us, d_vs)
+    d_outs, d_lens = eng.difference_pairs(d_us, d_vs)
+    for i in range(len(us)):
+        assert to_np(m_outs[i][:m_lens[i]]).tolist() == \
+            orc.merge_sorted([us[i], vs[i]]).tolist(), f"merge pair {i}"
+        assert to_np(d_outs[i][:d_lens[i]]).tolist() == \
+            orc.difference(us[i], vs[i]).tolist(), f"diff pair {i}"
+
+
+@pytest.mark.parametrize("k", [1, 2, 3, 8, 150])
+def test_kway_vs_oracle(eng, k):
+    rng = np.random.default_rng(SEED + k)
+    lists = [synth.gen_sorted_unique(rng, int(rng.integers(0, 20_000)), 60_000)
+             for _ in range(k)]
+    d_lists = [to_dev(x) for x in lists]
+    got_m = to_np(eng.merge_sorted(d_lists))
+    assert got_m.tolist() == orc.merge_sorted(lists).tolist()
+    got_i = to_np(eng.intersect_sorted(d_lists))
+    assert got_i.tolist() == orc.intersect_sorted(lists).tolist()
+
+
+def test_index_of_batch_vs_oracle(eng):
+    rng = np.random.default_rng(SEED)
+    u = synth.gen_sorted_unique(rng, 100_000, 1_000_000)
+    q = np.concatenate([u[::37], rng.integers(0, 1_000_000, 1000, dtype=np.uint64)])
+    got = eng.index_of_batch(to_dev(u), to_dev(q)).cpu().numpy()
+    want = [orc.index_of(u, int(x)) for x in q]
+    assert got.tolist() == want
+
+
+def test_extreme_values(eng):
+    """Boundary values incl. 2^63 crossing and UINT64_MAX (u64 compare, not i64)."""
+    u = np.array([0, 1, 2**32, 2**63 - 1, 2**63, 2**64 - 2, 2**64 - 1], dtype=np.uint64)
+    v = np.array([1, 2**63, 2**64 - 1], dtype=np.uint64)
+    outs, lens = eng.intersect_pairs([to_dev(u)], [to_dev(v)])
+    assert to_np(outs[0][:lens[0]]).tolist() == orc.intersect_with(u, v).tolist()
+    m_outs, m_lens = eng.merge_pairs([to_dev(u)], [to_dev(v)])
+    assert to_np(m_outs[0][:m_lens[0]]).tolist() == orc.merge_sorted([u, v]).tolist()
+    d_outs, d_lens = eng.difference_pairs([to_dev(u)], [to_dev(v)])
+    assert to_np(d_outs[0][:d_lens[0]]).tolist() == orc.difference(u, v).tolist()
+
+
+# ---------- packed (codec) path ----------
+
+@pytest.mark.parametrize("size,bs,m", [
+    (0, 256, 100), (1, 256, 100), (300, 10, 300), (5000, 0, 500),
+    (200_000, 256, 50_000), (2_000_000, 256, 1_000_000),
+])
+def test_packed_intersect_vs_oracle(eng, size, bs, m):
+    rng = np.random.default_rng(SEED + size + bs)
+    pack_uids = synth.getuids_geometric(rng, max(size, 1))[:size]
+    pack_uids = np.unique(pack_uids)  # duplicate-free parity domain
+    # v: half sampled from the pack (matches), half random
+    take = rng.choice(pack_uids.size, size=min(m // 2, pack_uids.size),
+                      replace=False) if pack_uids.size else []
+    hi = int(pack_uids[-1]) + 1000 if pack_uids.size else 1000
+    v = np.unique(np.concatenate([
+        pack_uids[np.sort(take)] if pack_uids.size else np.empty(0, np.uint64),
+        rng.integers(0, hi, size=m // 2, dtype=np.uint64)]))
+
+    bases, nums, offs, blob, total = algo.encode_flat(pack_uids, bs)
+    if total == 0:
+        return
+    dp = eng.upload_pack(bases, nums, offs, blob, bs)
+    got = to_np(eng.intersect_packed(dp, 0, to_dev(v)))
+
+    opack = orc.Pack(pack_uids, bs)
+    want = orc.intersect_compressed_with(opack, 0, v)
+    assert got.tolist() == want.tolist()
+
+    # decode parity (codec.Decode)
+    got_dec = to_np(eng.decode_pack(dp, 0))
+    assert got_dec.tolist() == opack.decode(0).tolist()
+
+    # after/seek parity
+    if pack_uids.size > 10:
+        for after in [int(pack_uids[pack_uids.size // 3]),
+                      int(pack_uids[pack_uids.size // 3]) + 1,
+                      int(pack_uids[-1]) + 1]:
+            got_a = to_np(eng.intersect_packed(dp, after, to_dev(v)))
+            want_a = orc.intersect_compressed_with(opack, after, v)
+            assert got_a.tolist() == want_a.tolist(), f"after={after}"
+            got_d = to_np(eng.decode_pack(dp, after))
+            assert got_d.tolist() == opack.decode(after).tolist()
+
+
+def test_packed_32msb_splits(eng):
+    """Blocks forced by 32-MSB changes (codec.go:117) incl. huge bases."""
+    rng = np.random.default_rng(SEED)
+    big = [0xf000000000000000, 0xf00f000000000000, 0x00f00f0000000000,
+           0x000f0f0000000000, 0x0f0f0f0f00000000]
+    vals = [np.uint64(rng.integers(0, 2**32)) for _ in range(50)]
+    vals += [np.uint64(rng.integers(0, 2**32)) + np.uint64(big[rng.integers(0, 5)])
+             for _ in range(50)]
+    uids = np.unique(np.array(vals, dtype=np.uint64))
+    v = uids[::3].copy()
+    bases, nums, offs, blob, total = algo.encode_flat(uids, 256)
+    dp = eng.upload_pack(bases, nums, offs, blob, 256)
+    got = to_np(eng.intersect_packed(dp, 0, to_dev(v)))
+    opack = orc.Pack(uids, 256)
+    assert got.tolist() == orc.intersect_compressed_with(opack, 0, v).tolist()
+    assert to_np(eng.decode_pack(dp, 0)).tolist() == uids.tolist()
+
+
+def test_host_packed_api(eng):
+    """ua_intersect_packed host-pointer convenience (the cgo surface)."""
+    import ctypes as Ct
+    from dgraph_amd import _lib
+    rng = np.random.default_rng(SEED)
+    uids = synth.getuids_geometric(rng, 10_000)
+    uids = np.unique(uids)
+    v = uids[::7].copy()
+    # build host ua_pack via ua_encode
+    h = Ct.c_void_p()
+    u_arr = np.ascontiguousarray(uids)
+    _lib.check(_lib.lib().ua_encode(
+        u_arr.ctypes.data_as(Ct.POINTER(Ct.c_uint64)), u_arr.size, 256, Ct.byref(h)))
+    try:
+        view = _lib.lib().ua_owned_pack_view(h)
+        out = np.empty(uids.size, dtype=np.uint64)
+        out_n = Ct.c_uint64()
+        _lib.check(_lib.lib().ua_intersect_packed(
+            eng._ctx, view, 0, v.ctypes.data_as(Ct.POINTER(Ct.c_uint64)), v.size,
+            out.ctypes.data_as(Ct.POINTER(Ct.c_uint64)), Ct.byref(out_n)))
+        opack = orc.Pack(uids, 256)
+        assert out[:out_n.value].tolist() == \
+            orc.intersect_compressed_with(opack, 0, v).tolist()
+    finally:
+        _lib.lib().ua_owned_pack_free(h)
