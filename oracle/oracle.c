@@ -341,8 +341,17 @@ void orc_dec_seek_to_block(orc_dec *d, uint64_t uid, int whence) {
     }
     const orc_pack *pack = d->pack;
     size_t nb = pack->n_blocks;
-    if (prev > 0 && (size_t)prev < nb && uid < pack->blocks[prev].base) prev = 0;
-    if ((size_t)prev >= nb) prev = 0; /* guard (Go would slice past end) */
+    if ((size_t)prev >= nb) {
+        /* Decoder exhausted (blockIdx == len(Blocks)).  The reference PANICS
+         * here — index out of range at codec.go:230 Blocks[prevBlockIdx] —
+         * reachable via IntersectCompressedWithBin when afterUID > every
+         * pack uid.  Result is undefined upstream; both this oracle and the
+         * GPU engine define it as the mathematically consistent empty set
+         * ({x in pack : x >= after} ∩ v = ∅). */
+        d->n_uids = 0;
+        return;
+    }
+    if (prev > 0 && uid < pack->blocks[prev].base) prev = 0;
 
     /* sort.Search over Blocks[prev:] (codec.go:245) */
     size_t cnt = nb - (size_t)prev;
