@@ -126,31 +126,43 @@ __global__ __launch_bounds__(UA_BLOCK) void k_partition(
 enum { OP_INTERSECT = 0, OP_UNION = 1, OP_DIFF = 2 };
 enum { MODE_STAGE = 0, MODE_COUNT = 1, MODE_WRITE = 2 };
 
+/* One pass over the thread's merge-path segment.  Emissions are recorded in
+ * a statically-indexed register array (em[s], flag bit s) so the walk runs
+ * ONCE: count + payload together, no re-walk, no scratch spill (guide §5.4
+ * rule 20: runtime-indexed local arrays go to scratch; per-step static
+ * indices do not). */
 template <int OP>
 __device__ __forceinline__ int tile_walk(const u64 *As, int alen, const u64 *Bs, int blen,
                                          u64 a_before, bool has_ab, bool has_bn,
-                                         int s0, int s1, int i0, u64 *emit) {
+                                         int s0, int s1, int i0,
+                                         u64 (&em)[UA_WPT], u32 &flags) {
     int i = i0, j = s0 - i0;
     int cnt = 0;
-    for (int s = s0; s < s1; s++) {
-        if (i >= alen && j >= blen) break;
+    flags = 0;
+    int steps = s1 - s0;
+#pragma unroll
+    for (int s = 0; s < UA_WPT; s++) {
+        if (s >= steps || (i >= alen && j >= blen)) break;
         bool takeA = (i < alen) && (j >= blen || As[i] <= Bs[j]);
         if (takeA) {
             u64 av = As[i];
             if (OP == OP_INTERSECT) {
                 bool match = (j < blen || has_bn) && (av == Bs[j]);
                 if (match) {
-                    if (emit) emit[cnt] = av;
+                    em[s] = av;
+                    flags |= 1u << s;
                     cnt++;
                 }
             } else if (OP == OP_DIFF) {
                 bool match = (j < blen || has_bn) && (av == Bs[j]);
                 if (!match) {
-                    if (emit) emit[cnt] = av;
+                    em[s] = av;
+                    flags |= 1u << s;
                     cnt++;
                 }
             } else { /* UNION: A always emits */
-                if (emit) emit[cnt] = av;
+                em[s] = av;
+                flags |= 1u << s;
                 cnt++;
             }
             i++;
@@ -160,7 +172,8 @@ __device__ __forceinline__ int tile_walk(const u64 *As, int alen, const u64 *Bs,
                 u64 ap = (i > 0) ? As[i - 1] : a_before;
                 bool dup = (i > 0 || has_ab) && (bv == ap);
                 if (!dup) {
-                    if (emit) emit[cnt] = bv;
+                    em[s] = bv;
+                    flags |= 1u << s;
                     cnt++;
                 }
             }
@@ -170,14 +183,60 @@ __device__ __forceinline__ int tile_walk(const u64 *As, int alen, const u64 *Bs,
     return cnt;
 }
 
+/* block-wide exclusive scan of per-thread counts: wave __shfl scan + one
+ * cross-wave combine — 1 barrier instead of Hillis-Steele's 17 */
+__device__ __forceinline__ void d_block_scan(int tid, u32 cnt, u32 *wsum,
+                                             u32 &excl, u32 &total) {
+    int lane = tid & 63, wv = tid >> 6;
+    u32 incl = cnt;
+#pragma unroll
+    for (int o = 1; o < 64; o <<= 1) {
+        u32 x = __shfl_up(incl, o);
+        if (lane >= o) incl += x;
+    }
+    if (lane == 63) wsum[wv] = incl;
+    __syncthreads();
+    u32 wbase = 0;
+#pragma unroll
+    for (int w = 0; w < UA_BLOCK / 64; w++) {
+        u32 s = wsum[w];
+        if (w < wv) wbase += s;
+    }
+    total = wsum[0] + wsum[1] + wsum[2] + wsum[3];
+    excl = wbase + incl - cnt;
+}
+
+/* cooperative global->LDS fill, 16-B vectorized on the aligned body
+ * (8-B/lane loads cap ~60% of the dwordx4 HBM rate — guide §2/G13) */
+__device__ __forceinline__ void d_fill_lds(u64 *dst, const u64 *__restrict__ src,
+                                           int len, int tid) {
+    int head = (int)(((uintptr_t)src >> 3) & 1); /* 1 if src ≡ 8 (mod 16) */
+    if (head > len) head = len;
+    if (tid == 0 && head) dst[0] = src[0];
+    int nvec = (len - head) >> 1;
+    const ulonglong2 *vs = (const ulonglong2 *)(src + head);
+    if ((((uintptr_t)(dst + head)) & 15) == 0) {
+        ulonglong2 *vd = (ulonglong2 *)(dst + head);
+        for (int i = tid; i < nvec; i += UA_BLOCK) vd[i] = vs[i];
+    } else {
+        for (int i = tid; i < nvec; i += UA_BLOCK) {
+            ulonglong2 x = vs[i];
+            dst[head + 2 * i] = x.x;
+            dst[head + 2 * i + 1] = x.y;
+        }
+    }
+    for (int i = head + 2 * nvec + tid; i < len; i += UA_BLOCK) dst[i] = src[i];
+}
+
 template <int OP, int MODE>
 __global__ __launch_bounds__(UA_BLOCK) void k_tiles(
     const UaDesc *__restrict__ descs, const u32 *__restrict__ tile_pair,
     const u32 *__restrict__ tile_a0, u64 total_tiles,
     u64 *__restrict__ staging, u64 stage_stride, u32 *__restrict__ tile_cnt,
     const u64 *__restrict__ offs /* MODE_WRITE only */) {
-    __shared__ u64 smem[UA_TILE + 4];
-    __shared__ u32 scan[UA_BLOCK];
+    __shared__ __align__(16) u64 smem[UA_TILE + 4];
+    __shared__ u32 scan[UA_BLOCK / 64]; /* per-wave totals for d_block_scan */
+    __shared__ u64 s_abefore;
 
     u64 t = blockIdx.x;
     int tid = threadIdx.x;
@@ -193,15 +252,15 @@ __global__ __launch_bounds__(UA_BLOCK) void k_tiles(
     u32 b0 = (u32)(d0 - a0), b1 = (u32)(d1 - a1);
     int alen = (int)(a1 - a0), blen = (int)(b1 - b0);
 
-    u64 *As = smem + 1;
-    u64 *Bs = smem + 2 + alen;
+    u64 *As = smem;                          /* 16-B aligned */
+    u64 *Bs = smem + ((alen + 1) & ~1);      /* rounded up to even: 16-B aligned */
 
-    for (int i = tid; i < alen; i += UA_BLOCK) As[i] = d.u[a0 + i];
-    for (int i = tid; i < blen; i += UA_BLOCK) Bs[i] = d.v[b0 + i];
+    d_fill_lds(As, d.u + a0, alen, tid);
+    d_fill_lds(Bs, d.v + b0, blen, tid);
     bool has_ab = (a0 > 0);
     bool has_bn = ((u64)b1 < d.m);
     if (tid == 0) {
-        smem[0] = has_ab ? d.u[a0 - 1] : 0;
+        s_abefore = has_ab ? d.u[a0 - 1] : 0;
         Bs[blen] = has_bn ? d.v[b1] : 0;
     }
     __syncthreads();
@@ -212,21 +271,15 @@ __global__ __launch_bounds__(UA_BLOCK) void k_tiles(
     if (s0 > tilelen) s0 = tilelen;
     if (s1 > tilelen) s1 = tilelen;
     int i0 = d_merge_path_lds(As, alen, Bs, blen, s0);
-    u64 a_before = smem[0];
+    u64 a_before = s_abefore;
 
-    int cnt = tile_walk<OP>(As, alen, Bs, blen, a_before, has_ab, has_bn, s0, s1, i0, nullptr);
+    u64 em[UA_WPT];
+    u32 flags;
+    int cnt = tile_walk<OP>(As, alen, Bs, blen, a_before, has_ab, has_bn, s0, s1, i0,
+                            em, flags);
 
-    /* exclusive scan of per-thread counts */
-    scan[tid] = (u32)cnt;
-    __syncthreads();
-    for (int off = 1; off < UA_BLOCK; off <<= 1) {
-        u32 x = (tid >= off) ? scan[tid - off] : 0;
-        __syncthreads();
-        scan[tid] += x;
-        __syncthreads();
-    }
-    u32 total = scan[UA_BLOCK - 1];
-    u32 excl = scan[tid] - (u32)cnt;
+    u32 excl, total;
+    d_block_scan(tid, (u32)cnt, scan, excl, total);
 
     if (MODE == MODE_COUNT) {
         if (tid == 0) tile_cnt[t] = total;
@@ -241,7 +294,11 @@ __global__ __launch_bounds__(UA_BLOCK) void k_tiles(
         dst = d.out + (offs[t] - pair_base) + excl;
     }
     if (cnt > 0) {
-        tile_walk<OP>(As, alen, Bs, blen, a_before, has_ab, has_bn, s0, s1, i0, dst);
+        int k = 0;
+#pragma unroll
+        for (int s = 0; s < UA_WPT; s++) {
+            if (flags & (1u << s)) dst[k++] = em[s];
+        }
     }
 }
 
