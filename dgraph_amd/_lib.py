@@ -7,7 +7,8 @@ import ctypes as C
 import os
 
 _DIR = os.path.dirname(os.path.abspath(__file__))
-_LIB_PATH = os.path.join(_DIR, "libuidalgo.so")
+# UA_LIB_PATH overrides the library for perf-bisection builds (tools/ubench.py)
+_LIB_PATH = os.environ.get("UA_LIB_PATH", os.path.join(_DIR, "libuidalgo.so"))
 
 UA_OK = 0
 

@@ -40,6 +40,11 @@ typedef uint32_t u32;
 typedef uint16_t u16;
 typedef uint8_t u8;
 
+#ifndef UA_ABLATE
+#define UA_ABLATE 0 /* perf-bisection builds only (tools/ubench.py): 1 = fill-only,
+                     * 2 = walk-on-garbage (no fill), 3 = no scan/write-back */
+#endif
+
 #define UA_TILE 2048      /* merge-path elements per tile */
 #define UA_BLOCK 256      /* threads per workgroup */
 #define UA_WPT (UA_TILE / UA_BLOCK) /* path elements per thread */
@@ -140,44 +145,50 @@ __device__ __forceinline__ int tile_walk(const u64 *As, int alen, const u64 *Bs,
     int cnt = 0;
     flags = 0;
     int steps = s1 - s0;
+    int blen_ext = blen + (has_bn ? 1 : 0); /* Bs[blen] is the lookahead slot */
+    /* frontier values live in registers: ONE masked LDS reload per step
+     * instead of re-gathering As[i]/Bs[j] every comparison */
+    u64 a = (i < alen) ? As[i] : 0;
+    u64 b = (j < blen_ext) ? Bs[j] : 0;
+    u64 prev_a = (i > 0) ? As[i - 1] : a_before; /* union dedup lookback */
 #pragma unroll
     for (int s = 0; s < UA_WPT; s++) {
         if (s >= steps || (i >= alen && j >= blen)) break;
-        bool takeA = (i < alen) && (j >= blen || As[i] <= Bs[j]);
+        bool takeA = (i < alen) && (j >= blen || a <= b);
         if (takeA) {
-            u64 av = As[i];
             if (OP == OP_INTERSECT) {
-                bool match = (j < blen || has_bn) && (av == Bs[j]);
+                bool match = (j < blen_ext) && (a == b);
                 if (match) {
-                    em[s] = av;
+                    em[s] = a;
                     flags |= 1u << s;
                     cnt++;
                 }
             } else if (OP == OP_DIFF) {
-                bool match = (j < blen || has_bn) && (av == Bs[j]);
+                bool match = (j < blen_ext) && (a == b);
                 if (!match) {
-                    em[s] = av;
+                    em[s] = a;
                     flags |= 1u << s;
                     cnt++;
                 }
             } else { /* UNION: A always emits */
-                em[s] = av;
+                em[s] = a;
                 flags |= 1u << s;
                 cnt++;
             }
+            if (OP == OP_UNION) prev_a = a;
             i++;
+            if (i < alen) a = As[i];
         } else {
             if (OP == OP_UNION) {
-                u64 bv = Bs[j];
-                u64 ap = (i > 0) ? As[i - 1] : a_before;
-                bool dup = (i > 0 || has_ab) && (bv == ap);
+                bool dup = (i > 0 || has_ab) && (b == prev_a);
                 if (!dup) {
-                    em[s] = bv;
+                    em[s] = b;
                     flags |= 1u << s;
                     cnt++;
                 }
             }
             j++;
+            if (j < blen_ext) b = Bs[j];
         }
     }
     return cnt;
@@ -255,14 +266,16 @@ __global__ __launch_bounds__(UA_BLOCK) void k_tiles(
     u64 *As = smem;                          /* 16-B aligned */
     u64 *Bs = smem + ((alen + 1) & ~1);      /* rounded up to even: 16-B aligned */
 
-    d_fill_lds(As, d.u + a0, alen, tid);
-    d_fill_lds(Bs, d.v + b0, blen, tid);
     bool has_ab = (a0 > 0);
     bool has_bn = ((u64)b1 < d.m);
+#if UA_ABLATE != 2
+    d_fill_lds(As, d.u + a0, alen, tid);
+    d_fill_lds(Bs, d.v + b0, blen, tid);
     if (tid == 0) {
         s_abefore = has_ab ? d.u[a0 - 1] : 0;
         Bs[blen] = has_bn ? d.v[b1] : 0;
     }
+#endif
     __syncthreads();
 
     int tilelen = alen + blen;
@@ -270,6 +283,14 @@ __global__ __launch_bounds__(UA_BLOCK) void k_tiles(
     int s1 = s0 + UA_WPT;
     if (s0 > tilelen) s0 = tilelen;
     if (s1 > tilelen) s1 = tilelen;
+#if UA_ABLATE == 1 /* fill-only: keep the loads live, skip search+walk */
+    u64 ablate_x = As[tid] + Bs[tid & 127];
+    asm volatile("" ::"v"(ablate_x));
+    u64 em[UA_WPT];
+    u32 flags = 0;
+    int cnt = 0;
+    (void)s_abefore;
+#else
     int i0 = d_merge_path_lds(As, alen, Bs, blen, s0);
     u64 a_before = s_abefore;
 
@@ -277,6 +298,12 @@ __global__ __launch_bounds__(UA_BLOCK) void k_tiles(
     u32 flags;
     int cnt = tile_walk<OP>(As, alen, Bs, blen, a_before, has_ab, has_bn, s0, s1, i0,
                             em, flags);
+#endif
+#if UA_ABLATE == 3 /* fill+walk, skip scan/write-back */
+    asm volatile("" ::"v"(cnt), "v"(flags));
+    cnt = 0;
+    flags = 0;
+#endif
 
     u32 excl, total;
     d_block_scan(tid, (u32)cnt, scan, excl, total);
@@ -371,29 +398,35 @@ __global__ __launch_bounds__(UA_BLOCK) void k_pair_out(const u64 *__restrict__ o
 
 /* ==================== kernel: compaction ==================== */
 
+/* one wavefront per tile (4 tiles per workgroup): 4x fewer blocks to
+ * dispatch, and tiles with cnt==0 cost one load */
 __global__ __launch_bounds__(UA_BLOCK) void k_compact(
     const UaDesc *__restrict__ descs, const u32 *__restrict__ tile_pair,
     const u32 *__restrict__ tile_cnt, const u64 *__restrict__ offs,
-    const u64 *__restrict__ staging, u64 stage_stride) {
-    u64 t = blockIdx.x;
+    const u64 *__restrict__ staging, u64 stage_stride, u64 total_tiles) {
+    u64 t = (u64)blockIdx.x * 4 + (threadIdx.x >> 6);
+    int lane = threadIdx.x & 63;
+    if (t >= total_tiles) return;
     u32 cnt = tile_cnt[t];
     if (cnt == 0) return;
     u32 p = tile_pair[t];
     UaDesc d = descs[p];
     u64 *dst = d.out + (offs[t] - offs[d.tile_base]);
     const u64 *src = staging + t * stage_stride;
-    for (u32 i = threadIdx.x; i < cnt; i += UA_BLOCK) dst[i] = src[i];
+    for (u32 i = lane; i < cnt; i += 64) dst[i] = src[i];
 }
 
 __global__ __launch_bounds__(UA_BLOCK) void k_compact_flat(
     u64 *__restrict__ out, const u32 *__restrict__ cnts, const u64 *__restrict__ offs,
-    const u64 *__restrict__ staging, u64 stage_stride) {
-    u64 b = blockIdx.x;
+    const u64 *__restrict__ staging, u64 stage_stride, u64 n_blocks) {
+    u64 b = (u64)blockIdx.x * 4 + (threadIdx.x >> 6);
+    int lane = threadIdx.x & 63;
+    if (b >= n_blocks) return;
     u32 cnt = cnts[b];
     if (cnt == 0) return;
     u64 *dst = out + offs[b];
     const u64 *src = staging + b * stage_stride;
-    for (u32 i = threadIdx.x; i < cnt; i += UA_BLOCK) dst[i] = src[i];
+    for (u32 i = lane; i < cnt; i += 64) dst[i] = src[i];
 }
 
 /* ==================== kernel: batched IndexOf ==================== */
@@ -782,8 +815,10 @@ static int run_batch_locked(ua_ctx *c, const ua_dpair *pairs, int n_pairs,
                                total_tiles, (u64 *)nullptr, 0, d_tcnt, d_toff);
             HIP_TRY(hipEventRecord(c->ev[3], c->stream));
         } else {
-            hipLaunchKernelGGL(k_compact, dim3((u32)total_tiles), dim3(UA_BLOCK), 0, c->stream,
-                               d_descs, d_tpair, d_tcnt, d_toff, d_stage, stage_stride);
+            u64 cblk = (total_tiles + 3) / 4;
+            hipLaunchKernelGGL(k_compact, dim3((u32)cblk), dim3(UA_BLOCK), 0, c->stream,
+                               d_descs, d_tpair, d_tcnt, d_toff, d_stage, stage_stride,
+                               total_tiles);
         }
     } else {
         HIP_TRY(hipMemsetAsync(d_toff, 0, (total_tiles + 1) * sizeof(u64), c->stream));
@@ -985,8 +1020,8 @@ static int run_packed_locked(ua_ctx *c, const ua_dpack *pk, u64 after, const u64
     HIP_TRY(hipEventRecord(c->ev[1], c->stream));
 
     if ((rc = run_scan(c, d_cnt, nb + 1, d_off))) return rc;
-    hipLaunchKernelGGL(k_compact_flat, dim3((u32)nb), dim3(UA_BLOCK), 0, c->stream,
-                       out, d_cnt, d_off, d_stage, (u64)UA_MAX_BLOCK_UIDS);
+    hipLaunchKernelGGL(k_compact_flat, dim3((u32)((nb + 3) / 4)), dim3(UA_BLOCK), 0,
+                       c->stream, out, d_cnt, d_off, d_stage, (u64)UA_MAX_BLOCK_UIDS, nb);
     HIP_TRY(hipMemcpyAsync(out_n, d_off + nb, sizeof(u64), hipMemcpyDeviceToHost, c->stream));
     HIP_TRY(hipStreamSynchronize(c->stream));
     HIP_TRY(hipGetLastError());
