@@ -45,7 +45,9 @@ typedef uint8_t u8;
                      * 2 = walk-on-garbage (no fill), 3 = no scan/write-back */
 #endif
 
-#define UA_TILE 2048      /* merge-path elements per tile */
+#ifndef UA_TILE
+#define UA_TILE 2048
+#endif      /* merge-path elements per tile */
 #define UA_BLOCK 256      /* threads per workgroup */
 #define UA_WPT (UA_TILE / UA_BLOCK) /* path elements per thread */
 #define UA_SCAN_CHUNK 2048
@@ -103,9 +105,15 @@ struct alignas(16) UaDesc {
 
 /* ==================== kernel: tile partition ==================== */
 
+/* Two-level tile partition.  Phase 0: full-range diagonal search for every
+ * 8th tile of each pair (and records tile->pair for all tiles).  Phase 1:
+ * the other 7/8 of tiles search only inside the bracket their two coarse
+ * neighbours pin (merge path is monotone), a <=16k-element window that lands
+ * in L2 — ~8x less random HBM fetch than a flat per-tile search. */
+#define UA_COARSE 8
 __global__ __launch_bounds__(UA_BLOCK) void k_partition(
     const UaDesc *__restrict__ descs, const u64 *__restrict__ tb, int n_pairs,
-    u64 total_tiles, u32 *__restrict__ tile_pair, u32 *__restrict__ tile_a0) {
+    u64 total_tiles, u32 *__restrict__ tile_pair, u32 *__restrict__ tile_a0, int phase) {
     u64 t = (u64)blockIdx.x * UA_BLOCK + threadIdx.x;
     if (t >= total_tiles) return;
     /* find pair p with tb[p] <= t < tb[p+1] */
@@ -116,14 +124,34 @@ __global__ __launch_bounds__(UA_BLOCK) void k_partition(
         else hi = mid - 1;
     }
     int p = lo;
-    UaDesc d = descs[p];
     u64 lt = t - tb[p];
+    int is_coarse = ((lt & (UA_COARSE - 1)) == 0);
+    if (phase == 0) {
+        tile_pair[t] = (u32)p;
+        if (!is_coarse) return;
+    } else {
+        if (is_coarse) return;
+    }
+    UaDesc d = descs[p];
     u64 path = d.n + d.m;
     u64 diag = lt * UA_TILE;
     if (diag > path) diag = path;
-    u64 a0 = d_merge_path(d.u, d.n, d.v, d.m, diag);
-    tile_pair[t] = (u32)p;
-    tile_a0[t] = (u32)a0;
+    u64 slo = (diag > d.m) ? diag - d.m : 0;
+    u64 shi = diag < d.n ? diag : d.n;
+    if (phase == 1) {
+        u64 t_down = t - (lt & (UA_COARSE - 1));
+        u64 bl = tile_a0[t_down];
+        u64 t_up = t_down + UA_COARSE;
+        u64 bh = (t_up < tb[p + 1]) ? (u64)tile_a0[t_up] : d.n;
+        if (bl > slo) slo = bl;
+        if (bh < shi) shi = bh;
+    }
+    while (slo < shi) {
+        u64 mid = (slo + shi) >> 1;
+        if (d.u[mid] <= d.v[diag - 1 - mid]) slo = mid + 1;
+        else shi = mid;
+    }
+    tile_a0[t] = (u32)slo;
 }
 
 /* ==================== kernel: tile set-algebra ==================== */
@@ -788,7 +816,9 @@ static int run_batch_locked(ua_ctx *c, const ua_dpair *pairs, int n_pairs,
     if (total_tiles > 0) {
         u64 pblk = (total_tiles + UA_BLOCK - 1) / UA_BLOCK;
         hipLaunchKernelGGL(k_partition, dim3((u32)pblk), dim3(UA_BLOCK), 0, c->stream,
-                           d_descs, d_tb, n_pairs, total_tiles, d_tpair, d_ta0);
+                           d_descs, d_tb, n_pairs, total_tiles, d_tpair, d_ta0, 0);
+        hipLaunchKernelGGL(k_partition, dim3((u32)pblk), dim3(UA_BLOCK), 0, c->stream,
+                           d_descs, d_tb, n_pairs, total_tiles, d_tpair, d_ta0, 1);
 
         HIP_TRY(hipEventRecord(c->ev[0], c->stream));
         if (op == OP_INTERSECT) {
