@@ -43,13 +43,15 @@ def log(msg):
         print(msg, file=sys.stderr, flush=True)
 
 
-def load_traffic():
-    """Per-launch HBM bytes from a committed rocprofv3 PMC summary, if any."""
+def load_traffic(pairs):
+    """Per-launch HBM bytes from a committed rocprofv3 PMC summary, if any.
+    Only valid when the launch shape matches the profiled one."""
     path = os.path.join(ROOT, "profiles", "hbm_traffic.json")
     if os.path.exists(path):
         try:
             d = json.load(open(path))
-            return d.get("bytes_per_launch")
+            if f"{pairs} pairs/launch" in d.get("workload", ""):
+                return d.get("bytes_per_launch")
         except Exception:
             return None
     return None
@@ -165,7 +167,7 @@ def main():
             "peak": HBM_PEAK_GBS,
             "unit": "GB/s",
             "frac": round(achieved_gbs / HBM_PEAK_GBS, 4),
-            "traffic": load_traffic(),
+            "traffic": load_traffic(P),
             "kernel": "k_tiles<OP_INTERSECT,MODE_STAGE>",
             "kernel_ms_avg": round(stats["kernel_ms"] / stats["launches"], 4),
             "launches": stats["launches"],

@@ -92,6 +92,24 @@ __device__ __forceinline__ int d_merge_path_lds(const u64 *A, int n, const u64 *
     return lo;
 }
 
+__device__ __forceinline__ int d_lower_bound_i(const u64 *a, int n, u64 key) {
+    int lo = 0, hi = n;
+    while (lo < hi) {
+        int mid = (lo + hi) >> 1;
+        if (a[mid] >= key) hi = mid;
+        else lo = mid + 1;
+    }
+    return lo;
+}
+
+/* No scan-3 pass: the flat exclusive scan stays split as (chunk-local offs,
+ * chunk partials); every consumer computes the final offset with one extra
+ * L2 load — saves a full read+write sweep of offs plus a launch. */
+__device__ __forceinline__ u64 d_off(const u64 *__restrict__ offs,
+                                     const u64 *__restrict__ partials, u64 x) {
+    return offs[x] + partials[x / UA_SCAN_CHUNK];
+}
+
 /* ==================== batch descriptors ==================== */
 
 struct alignas(16) UaDesc {
@@ -245,6 +263,55 @@ __device__ __forceinline__ void d_block_scan(int tid, u32 cnt, u32 *wsum,
     excl = wbase + incl - cnt;
 }
 
+/* Advance j to the first index in [j, bext) with Bs[idx] >= a: gallop then
+ * binary — O(log gap) LDS probes, no serial per-element dependence. */
+__device__ __forceinline__ int d_advance(const u64 *Bs, int bext, int j, u64 a) {
+    if (j >= bext || Bs[j] >= a) return j;
+    int st = 1;
+    while (j + st < bext && Bs[j + st] < a) st <<= 1;
+    int lo = j + (st >> 1) + 1; /* Bs[j + st/2] < a */
+    int hi = j + st;
+    if (hi > bext) hi = bext;
+    while (lo < hi) {
+        int mid = (lo + hi) >> 1;
+        if (Bs[mid] >= a) hi = mid;
+        else lo = mid + 1;
+    }
+    return lo;
+}
+
+/* A-indexed tile pass for INTERSECT/DIFF: threads take contiguous A-index
+ * chunks (output order = A order since only A elements emit), and find each
+ * A element in the tile's B range by monotone galloping search.  ~2x fewer
+ * instructions than the merge walk (the kernel is issue-bound), and no
+ * per-thread diagonal search. */
+template <int OP>
+__device__ __forceinline__ int tile_search(const u64 *As, int alen, const u64 *Bs,
+                                           int blen, bool has_bn, int tid,
+                                           u64 (&em)[UA_WPT], u32 &flags) {
+    int lo = (int)(((long)tid * alen) / UA_BLOCK);
+    int hi = (int)(((long)(tid + 1) * alen) / UA_BLOCK);
+    int bext = blen + (has_bn ? 1 : 0);
+    int cnt = 0;
+    flags = 0;
+    if (lo >= hi) return 0;
+    int j = d_lower_bound_i(Bs, bext, As[lo]);
+#pragma unroll
+    for (int s = 0; s < UA_WPT; s++) {
+        if (lo + s >= hi) break;
+        u64 a = As[lo + s];
+        j = d_advance(Bs, bext, j, a);
+        bool found = (j < bext) && (Bs[j] == a);
+        bool keep = (OP == OP_INTERSECT) ? found : !found;
+        if (keep) {
+            em[s] = a;
+            flags |= 1u << s;
+            cnt++;
+        }
+    }
+    return cnt;
+}
+
 /* cooperative global->LDS fill, 16-B vectorized on the aligned body
  * (8-B/lane loads cap ~60% of the dwordx4 HBM rate — guide §2/G13) */
 __device__ __forceinline__ void d_fill_lds(u64 *dst, const u64 *__restrict__ src,
@@ -272,7 +339,7 @@ __global__ __launch_bounds__(UA_BLOCK) void k_tiles(
     const UaDesc *__restrict__ descs, const u32 *__restrict__ tile_pair,
     const u32 *__restrict__ tile_a0, u64 total_tiles,
     u64 *__restrict__ staging, u64 stage_stride, u32 *__restrict__ tile_cnt,
-    const u64 *__restrict__ offs /* MODE_WRITE only */) {
+    const u64 *__restrict__ offs, const u64 *__restrict__ partials /* MODE_WRITE only */) {
     __shared__ __align__(16) u64 smem[UA_TILE + 4];
     __shared__ u32 scan[UA_BLOCK / 64]; /* per-wave totals for d_block_scan */
     __shared__ u64 s_abefore;
@@ -306,11 +373,9 @@ __global__ __launch_bounds__(UA_BLOCK) void k_tiles(
 #endif
     __syncthreads();
 
-    int tilelen = alen + blen;
-    int s0 = tid * UA_WPT;
-    int s1 = s0 + UA_WPT;
-    if (s0 > tilelen) s0 = tilelen;
-    if (s1 > tilelen) s1 = tilelen;
+#ifndef UA_SEARCH
+#define UA_SEARCH 0 /* 1 = A-indexed gallop search for intersect/diff; measured 3.2 vs 3.45 TB/s for the merge walk on cfg2 - kept for skewed-ratio experiments */
+#endif
 #if UA_ABLATE == 1 /* fill-only: keep the loads live, skip search+walk */
     u64 ablate_x = As[tid] + Bs[tid & 127];
     asm volatile("" ::"v"(ablate_x));
@@ -319,13 +384,22 @@ __global__ __launch_bounds__(UA_BLOCK) void k_tiles(
     int cnt = 0;
     (void)s_abefore;
 #else
-    int i0 = d_merge_path_lds(As, alen, Bs, blen, s0);
-    u64 a_before = s_abefore;
-
     u64 em[UA_WPT];
     u32 flags;
-    int cnt = tile_walk<OP>(As, alen, Bs, blen, a_before, has_ab, has_bn, s0, s1, i0,
+    int cnt;
+    if (UA_SEARCH && OP != OP_UNION) {
+        cnt = tile_search<OP>(As, alen, Bs, blen, has_bn, tid, em, flags);
+    } else {
+        int tilelen = alen + blen;
+        int s0 = tid * UA_WPT;
+        int s1 = s0 + UA_WPT;
+        if (s0 > tilelen) s0 = tilelen;
+        if (s1 > tilelen) s1 = tilelen;
+        int i0 = d_merge_path_lds(As, alen, Bs, blen, s0);
+        u64 a_before = s_abefore;
+        cnt = tile_walk<OP>(As, alen, Bs, blen, a_before, has_ab, has_bn, s0, s1, i0,
                             em, flags);
+    }
 #endif
 #if UA_ABLATE == 3 /* fill+walk, skip scan/write-back */
     asm volatile("" ::"v"(cnt), "v"(flags));
@@ -344,9 +418,9 @@ __global__ __launch_bounds__(UA_BLOCK) void k_tiles(
     if (MODE == MODE_STAGE) {
         dst = staging + t * stage_stride + excl;
         if (tid == 0) tile_cnt[t] = total;
-    } else { /* MODE_WRITE: offs is the flat exclusive scan of tile counts */
-        u64 pair_base = offs[d.tile_base];
-        dst = d.out + (offs[t] - pair_base) + excl;
+    } else { /* MODE_WRITE: (offs, partials) is the split flat scan */
+        dst = d.out +
+              (d_off(offs, partials, t) - d_off(offs, partials, d.tile_base)) + excl;
     }
     if (cnt > 0) {
         int k = 0;
@@ -411,50 +485,53 @@ __global__ __launch_bounds__(UA_BLOCK) void k_scan2(u64 *__restrict__ partials, 
     }
 }
 
-__global__ __launch_bounds__(UA_BLOCK) void k_scan3(u64 *__restrict__ offs, u64 n,
-                                                    const u64 *__restrict__ partials) {
-    u64 i = (u64)blockIdx.x * UA_BLOCK + threadIdx.x;
-    if (i < n) offs[i] += partials[i / UA_SCAN_CHUNK];
-}
-
 __global__ __launch_bounds__(UA_BLOCK) void k_pair_out(const u64 *__restrict__ offs,
+                                                       const u64 *__restrict__ partials,
                                                        const u64 *__restrict__ tb, int n_pairs,
                                                        u64 *__restrict__ pout) {
     int p = blockIdx.x * UA_BLOCK + threadIdx.x;
-    if (p < n_pairs) pout[p] = offs[tb[p + 1]] - offs[tb[p]];
+    if (p < n_pairs) pout[p] = d_off(offs, partials, tb[p + 1]) - d_off(offs, partials, tb[p]);
 }
 
 /* ==================== kernel: compaction ==================== */
 
-/* one wavefront per tile (4 tiles per workgroup): 4x fewer blocks to
+/* one wavefront per tile, 4 tiles per wave, 16 per workgroup: few blocks to
  * dispatch, and tiles with cnt==0 cost one load */
 __global__ __launch_bounds__(UA_BLOCK) void k_compact(
     const UaDesc *__restrict__ descs, const u32 *__restrict__ tile_pair,
     const u32 *__restrict__ tile_cnt, const u64 *__restrict__ offs,
-    const u64 *__restrict__ staging, u64 stage_stride, u64 total_tiles) {
-    u64 t = (u64)blockIdx.x * 4 + (threadIdx.x >> 6);
+    const u64 *__restrict__ partials, const u64 *__restrict__ staging,
+    u64 stage_stride, u64 total_tiles) {
+    u64 base = ((u64)blockIdx.x * 4 + (threadIdx.x >> 6)) * 4;
     int lane = threadIdx.x & 63;
-    if (t >= total_tiles) return;
-    u32 cnt = tile_cnt[t];
-    if (cnt == 0) return;
-    u32 p = tile_pair[t];
-    UaDesc d = descs[p];
-    u64 *dst = d.out + (offs[t] - offs[d.tile_base]);
-    const u64 *src = staging + t * stage_stride;
-    for (u32 i = lane; i < cnt; i += 64) dst[i] = src[i];
+    for (int q = 0; q < 4; q++) {
+        u64 t = base + q;
+        if (t >= total_tiles) return;
+        u32 cnt = tile_cnt[t];
+        if (cnt == 0) continue;
+        u32 p = tile_pair[t];
+        UaDesc d = descs[p];
+        u64 *dst = d.out + (d_off(offs, partials, t) - d_off(offs, partials, d.tile_base));
+        const u64 *src = staging + t * stage_stride;
+        for (u32 i = lane; i < cnt; i += 64) dst[i] = src[i];
+    }
 }
 
 __global__ __launch_bounds__(UA_BLOCK) void k_compact_flat(
     u64 *__restrict__ out, const u32 *__restrict__ cnts, const u64 *__restrict__ offs,
-    const u64 *__restrict__ staging, u64 stage_stride, u64 n_blocks) {
-    u64 b = (u64)blockIdx.x * 4 + (threadIdx.x >> 6);
+    const u64 *__restrict__ partials, const u64 *__restrict__ staging,
+    u64 stage_stride, u64 n_blocks) {
+    u64 base = ((u64)blockIdx.x * 4 + (threadIdx.x >> 6)) * 4;
     int lane = threadIdx.x & 63;
-    if (b >= n_blocks) return;
-    u32 cnt = cnts[b];
-    if (cnt == 0) return;
-    u64 *dst = out + offs[b];
-    const u64 *src = staging + b * stage_stride;
-    for (u32 i = lane; i < cnt; i += 64) dst[i] = src[i];
+    for (int q = 0; q < 4; q++) {
+        u64 b = base + q;
+        if (b >= n_blocks) return;
+        u32 cnt = cnts[b];
+        if (cnt == 0) continue;
+        u64 *dst = out + d_off(offs, partials, b);
+        const u64 *src = staging + b * stage_stride;
+        for (u32 i = lane; i < cnt; i += 64) dst[i] = src[i];
+    }
 }
 
 /* ==================== kernel: batched IndexOf ==================== */
@@ -747,6 +824,8 @@ extern "C" int ua_stats_get(ua_ctx *c, uint64_t *n_launches, double *kernel_ms,
 
 /* ---- flat scan helper: cnt u32[n] (+1 zero sentinel at n-1 position
  * provided by caller) -> offs u64[n] exclusive scan ---- */
+/* Split flat scan: offs_dev gets chunk-local exclusive offsets, WS_PARTIAL
+ * gets the scanned chunk partials; consumers combine via d_off(). */
 static int run_scan(ua_ctx *c, const u32 *cnt_dev, u64 n, u64 *offs_dev) {
     u64 nchunks = (n + UA_SCAN_CHUNK - 1) / UA_SCAN_CHUNK;
     int rc = ws_reserve(c, WS_PARTIAL, (nchunks + 1) * sizeof(u64));
@@ -755,9 +834,6 @@ static int run_scan(ua_ctx *c, const u32 *cnt_dev, u64 n, u64 *offs_dev) {
     hipLaunchKernelGGL(k_scan1, dim3((u32)nchunks), dim3(UA_BLOCK), 0, c->stream,
                        cnt_dev, n, offs_dev, partials);
     hipLaunchKernelGGL(k_scan2, dim3(1), dim3(UA_BLOCK), 0, c->stream, partials, nchunks);
-    u64 nblk3 = (n + UA_BLOCK - 1) / UA_BLOCK;
-    hipLaunchKernelGGL(k_scan3, dim3((u32)nblk3), dim3(UA_BLOCK), 0, c->stream,
-                       offs_dev, n, partials);
     return UA_OK;
 }
 
@@ -782,8 +858,9 @@ static int run_batch_locked(ua_ctx *c, const ua_dpair *pairs, int n_pairs,
     tb[n_pairs] = total_tiles;
 
     int rc;
-    if ((rc = ws_reserve(c, WS_DESC, descs.size() * sizeof(UaDesc)))) return rc;
-    if ((rc = ws_reserve(c, WS_TB, tb.size() * sizeof(u64)))) return rc;
+    /* descs and tb share one allocation and one upload */
+    size_t descs_bytes = descs.size() * sizeof(UaDesc);
+    if ((rc = ws_reserve(c, WS_DESC, descs_bytes + tb.size() * sizeof(u64)))) return rc;
     if ((rc = ws_reserve(c, WS_TPAIR, (total_tiles + 1) * sizeof(u32)))) return rc;
     if ((rc = ws_reserve(c, WS_TA0, (total_tiles + 1) * sizeof(u32)))) return rc;
     if ((rc = ws_reserve(c, WS_TCNT, (total_tiles + 1) * sizeof(u32)))) return rc;
@@ -798,7 +875,7 @@ static int run_batch_locked(ua_ctx *c, const ua_dpair *pairs, int n_pairs,
     }
 
     UaDesc *d_descs = (UaDesc *)c->ws[WS_DESC];
-    u64 *d_tb = (u64 *)c->ws[WS_TB];
+    u64 *d_tb = (u64 *)((u8 *)c->ws[WS_DESC] + descs_bytes);
     u32 *d_tpair = (u32 *)c->ws[WS_TPAIR];
     u32 *d_ta0 = (u32 *)c->ws[WS_TA0];
     u32 *d_tcnt = (u32 *)c->ws[WS_TCNT];
@@ -806,9 +883,10 @@ static int run_batch_locked(ua_ctx *c, const ua_dpair *pairs, int n_pairs,
     u64 *d_pout = (u64 *)c->ws[WS_POUT];
     u64 *d_stage = (u64 *)c->ws[WS_STAGE];
 
-    HIP_TRY(hipMemcpyAsync(d_descs, descs.data(), descs.size() * sizeof(UaDesc),
-                           hipMemcpyHostToDevice, c->stream));
-    HIP_TRY(hipMemcpyAsync(d_tb, tb.data(), tb.size() * sizeof(u64),
+    std::vector<u8> hostbuf(descs_bytes + tb.size() * sizeof(u64));
+    memcpy(hostbuf.data(), descs.data(), descs_bytes);
+    memcpy(hostbuf.data() + descs_bytes, tb.data(), tb.size() * sizeof(u64));
+    HIP_TRY(hipMemcpyAsync(d_descs, hostbuf.data(), hostbuf.size(),
                            hipMemcpyHostToDevice, c->stream));
     /* zero sentinel so offs[total_tiles] = total output */
     HIP_TRY(hipMemsetAsync(d_tcnt + total_tiles, 0, sizeof(u32), c->stream));
@@ -824,39 +902,45 @@ static int run_batch_locked(ua_ctx *c, const ua_dpair *pairs, int n_pairs,
         if (op == OP_INTERSECT) {
             hipLaunchKernelGGL((k_tiles<OP_INTERSECT, MODE_STAGE>), dim3((u32)total_tiles),
                                dim3(UA_BLOCK), 0, c->stream, d_descs, d_tpair, d_ta0,
-                               total_tiles, d_stage, stage_stride, d_tcnt, (u64 *)nullptr);
+                               total_tiles, d_stage, stage_stride, d_tcnt, (u64 *)nullptr,
+                               (u64 *)nullptr);
         } else if (op == OP_DIFF) {
             hipLaunchKernelGGL((k_tiles<OP_DIFF, MODE_STAGE>), dim3((u32)total_tiles),
                                dim3(UA_BLOCK), 0, c->stream, d_descs, d_tpair, d_ta0,
-                               total_tiles, d_stage, stage_stride, d_tcnt, (u64 *)nullptr);
+                               total_tiles, d_stage, stage_stride, d_tcnt, (u64 *)nullptr,
+                               (u64 *)nullptr);
         } else {
             hipLaunchKernelGGL((k_tiles<OP_UNION, MODE_COUNT>), dim3((u32)total_tiles),
                                dim3(UA_BLOCK), 0, c->stream, d_descs, d_tpair, d_ta0,
-                               total_tiles, (u64 *)nullptr, 0, d_tcnt, (u64 *)nullptr);
+                               total_tiles, (u64 *)nullptr, 0, d_tcnt, (u64 *)nullptr,
+                               (u64 *)nullptr);
         }
         HIP_TRY(hipEventRecord(c->ev[1], c->stream));
 
         if ((rc = run_scan(c, d_tcnt, total_tiles + 1, d_toff))) return rc;
+        u64 *d_part = (u64 *)c->ws[WS_PARTIAL];
 
         if (op == OP_UNION) {
             HIP_TRY(hipEventRecord(c->ev[2], c->stream));
             hipLaunchKernelGGL((k_tiles<OP_UNION, MODE_WRITE>), dim3((u32)total_tiles),
                                dim3(UA_BLOCK), 0, c->stream, d_descs, d_tpair, d_ta0,
-                               total_tiles, (u64 *)nullptr, 0, d_tcnt, d_toff);
+                               total_tiles, (u64 *)nullptr, 0, d_tcnt, d_toff, d_part);
             HIP_TRY(hipEventRecord(c->ev[3], c->stream));
         } else {
-            u64 cblk = (total_tiles + 3) / 4;
+            u64 cblk = (total_tiles + 15) / 16;
             hipLaunchKernelGGL(k_compact, dim3((u32)cblk), dim3(UA_BLOCK), 0, c->stream,
-                               d_descs, d_tpair, d_tcnt, d_toff, d_stage, stage_stride,
-                               total_tiles);
+                               d_descs, d_tpair, d_tcnt, d_toff, d_part, d_stage,
+                               stage_stride, total_tiles);
         }
     } else {
+        if ((rc = ws_reserve(c, WS_PARTIAL, sizeof(u64)))) return rc;
         HIP_TRY(hipMemsetAsync(d_toff, 0, (total_tiles + 1) * sizeof(u64), c->stream));
+        HIP_TRY(hipMemsetAsync(c->ws[WS_PARTIAL], 0, sizeof(u64), c->stream));
     }
 
     u64 poutblk = ((u64)n_pairs + UA_BLOCK - 1) / UA_BLOCK;
     hipLaunchKernelGGL(k_pair_out, dim3((u32)poutblk), dim3(UA_BLOCK), 0, c->stream,
-                       d_toff, d_tb, n_pairs, d_pout);
+                       d_toff, (const u64 *)c->ws[WS_PARTIAL], d_tb, n_pairs, d_pout);
     HIP_TRY(hipMemcpyAsync(out_lens, d_pout, (size_t)n_pairs * sizeof(u64),
                            hipMemcpyDeviceToHost, c->stream));
     HIP_TRY(hipStreamSynchronize(c->stream));
@@ -1050,11 +1134,18 @@ static int run_packed_locked(ua_ctx *c, const ua_dpack *pk, u64 after, const u64
     HIP_TRY(hipEventRecord(c->ev[1], c->stream));
 
     if ((rc = run_scan(c, d_cnt, nb + 1, d_off))) return rc;
-    hipLaunchKernelGGL(k_compact_flat, dim3((u32)((nb + 3) / 4)), dim3(UA_BLOCK), 0,
-                       c->stream, out, d_cnt, d_off, d_stage, (u64)UA_MAX_BLOCK_UIDS, nb);
-    HIP_TRY(hipMemcpyAsync(out_n, d_off + nb, sizeof(u64), hipMemcpyDeviceToHost, c->stream));
+    u64 *d_part = (u64 *)c->ws[WS_PARTIAL];
+    hipLaunchKernelGGL(k_compact_flat, dim3((u32)((nb + 15) / 16)), dim3(UA_BLOCK), 0,
+                       c->stream, out, d_cnt, d_off, d_part, d_stage,
+                       (u64)UA_MAX_BLOCK_UIDS, nb);
+    u64 off_nb = 0, part_nb = 0;
+    HIP_TRY(hipMemcpyAsync(&off_nb, d_off + nb, sizeof(u64), hipMemcpyDeviceToHost,
+                           c->stream));
+    HIP_TRY(hipMemcpyAsync(&part_nb, d_part + nb / UA_SCAN_CHUNK, sizeof(u64),
+                           hipMemcpyDeviceToHost, c->stream));
     HIP_TRY(hipStreamSynchronize(c->stream));
     HIP_TRY(hipGetLastError());
+    *out_n = off_nb + part_nb;
 
     float ms = 0.f;
     HIP_TRY(hipEventElapsedTime(&ms, c->ev[0], c->ev[1]));
