@@ -94,6 +94,9 @@ def lib():
                                               _u64, C.c_void_p, _u64p]
         L.ua_decode_dev.argtypes = [C.c_void_p, C.POINTER(UaDPack), _u64, C.c_void_p, _u64p]
         L.ua_encode.argtypes = [_u64p, _u64, C.c_uint32, _voidpp]
+        L.ua_encode_dev.argtypes = [C.c_void_p, C.c_void_p, _u64, C.c_uint32,
+                                    C.c_void_p, C.c_void_p, C.c_void_p, C.c_void_p,
+                                    _u64p, _u64p]
         L.ua_owned_pack_view.restype = C.POINTER(UaPack)
         L.ua_owned_pack_view.argtypes = [C.c_void_p]
         L.ua_owned_pack_free.argtypes = [C.c_void_p]

@@ -179,6 +179,50 @@ class Engine:
             v.numel(), C.c_void_p(out.data_ptr()), C.byref(out_n)))
         return out[:out_n.value]
 
+    def encode_dev(self, uids, block_size=256):
+        """codec.Encode on the GPU (codec.go:393): sorted device int64 (u64
+        bits) -> DPack resident in HBM.  Byte-identical to the reference
+        group-varint block format."""
+        import torch
+        n = uids.numel()
+        dev = uids.device
+        mb = max(n, 1)
+        bases = torch.empty(mb, dtype=torch.int64, device=dev)
+        nums = torch.empty(mb, dtype=torch.int32, device=dev)
+        offs = torch.empty(mb + 1, dtype=torch.int64, device=dev)
+        blob = torch.empty(6 * mb + 24, dtype=torch.int8, device=dev)
+        nb = _u64()
+        db = _u64()
+        check(lib().ua_encode_dev(
+            self._ctx, C.c_void_p(uids.data_ptr()), n, block_size,
+            C.c_void_p(bases.data_ptr()), C.c_void_p(nums.data_ptr()),
+            C.c_void_p(offs.data_ptr()), C.c_void_p(blob.data_ptr()),
+            C.byref(nb), C.byref(db)))
+        k = nb.value
+        return DPack(bases[:max(k, 1)], nums[:max(k, 1)], offs[:k + 1],
+                     blob[:max(db.value, 1)], int(block_size), n) if k else \
+            DPack(bases[:0], nums[:0], offs[:1], blob[:1], int(block_size), 0)
+
+    # ---- packed compositions (algo/packed.go) ----
+    def merge_sorted_packed(self, lists, block_size=256):
+        """algo.MergeSortedPacked (packed.go:222): dedup k-way union, packed
+        on-device."""
+        return self.encode_dev(self.merge_sorted(lists), block_size)
+
+    def intersect_sorted_packed(self, dpacks):
+        """algo.IntersectSortedPacked (packed.go:100): decode each pack,
+        fold-intersect smallest-first, re-pack (block size of the first
+        list, like the reference)."""
+        bs = dpacks[0].block_size if dpacks else 10
+        lists = [self.decode_pack(dp) for dp in dpacks]
+        return self.encode_dev(self.intersect_sorted(lists), bs)
+
+    def apply_filter_packed(self, dpack, mask_fn):
+        """algo.ApplyFilterPacked (packed.go:16): decode, boolean-mask
+        compaction, re-pack."""
+        dec = self.decode_pack(dpack)
+        return self.encode_dev(dec[mask_fn(dec)], dpack.block_size)
+
     def decode_pack(self, dpack, seek=0, out=None):
         """codec.Decode(pack, seek) on the GPU (codec.go:444)."""
         import torch

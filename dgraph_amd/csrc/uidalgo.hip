@@ -682,6 +682,134 @@ __global__ __launch_bounds__(UA_BLOCK) void k_packed(
     if (lane == 0) blk_cnt[b] = cnt;
 }
 
+/* ==================== kernels: GPU codec.Encode ====================
+ * Parallel restatement of codec.go:57-136 (packBlock/Add): block boundaries
+ * are 32-MSB changes (match32MSB :469) plus every block_size-th element
+ * within a 32-MSB run; each block's group-varint bytes are produced by one
+ * wavefront (lane g encodes group g; a wave shfl-scan of group byte-lengths
+ * places them).  Byte-identical to the reference format. */
+
+#define UA_ENC_STRIDE 1104 /* per-block staging bytes (64 groups x 17, padded) */
+
+__global__ __launch_bounds__(UA_BLOCK) void k_enc_msb_flags(
+    const u64 *__restrict__ uids, u64 n, u32 *__restrict__ flags) {
+    u64 i = (u64)blockIdx.x * UA_BLOCK + threadIdx.x;
+    if (i >= n) return;
+    flags[i] = (i == 0) || (((uids[i] ^ uids[i - 1]) >> 32) != 0);
+}
+
+/* run/block starts: out[rank(i)] = i for flagged i; thread n writes the
+ * total sentinel out[count] = n */
+__global__ __launch_bounds__(UA_BLOCK) void k_enc_mark_start(
+    const u32 *__restrict__ flags, const u64 *__restrict__ offs,
+    const u64 *__restrict__ partials, u64 n, u64 *__restrict__ out) {
+    u64 i = (u64)blockIdx.x * UA_BLOCK + threadIdx.x;
+    if (i > n) return;
+    if (i == n) {
+        out[d_off(offs, partials, n)] = n;
+        return;
+    }
+    if (flags[i]) out[d_off(offs, partials, i)] = i;
+}
+
+/* msb flags -> block flags (in place): also split every block_size-th
+ * element of each 32-MSB run (codec.go:117-126; blockSize 0 -> 1-uid blocks) */
+__global__ __launch_bounds__(UA_BLOCK) void k_enc_block_flags(
+    u32 *__restrict__ flags, const u64 *__restrict__ offs,
+    const u64 *__restrict__ partials, const u64 *__restrict__ run_start, u64 n,
+    u32 bs) {
+    u64 i = (u64)blockIdx.x * UA_BLOCK + threadIdx.x;
+    if (i >= n) return;
+    if (flags[i]) return; /* already a boundary */
+    u64 rid = d_off(offs, partials, i) + flags[i] - 1; /* inclusive rank - 1 */
+    u64 rs = run_start[rid];
+    u32 eff = bs ? bs : 1;
+    flags[i] = (((i - rs) % eff) == 0);
+}
+
+__global__ __launch_bounds__(UA_BLOCK) void k_encode_blocks(
+    const u64 *__restrict__ uids, const u64 *__restrict__ bstart, u64 nb,
+    u64 *__restrict__ bases, u32 *__restrict__ nums, u32 *__restrict__ blk_bytes,
+    u8 *__restrict__ stage) {
+    __shared__ u8 sbytes[UA_PKW][UA_ENC_STRIDE];
+    int wv = threadIdx.x >> 6;
+    int lane = threadIdx.x & 63;
+    u64 b = (u64)blockIdx.x * UA_PKW + wv;
+    bool active = (b < nb);
+    u64 s = 0, e = 0;
+    u32 num = 0, ng = 0;
+    if (active) {
+        s = bstart[b];
+        e = bstart[b + 1];
+        num = (u32)(e - s);
+        ng = (num > 1) ? ((num + 2) >> 2) : 1; /* >=1 group incl. pad-only (codec.go:76-96) */
+        if (lane == 0) {
+            bases[b] = uids[s];
+            nums[b] = num;
+        }
+    }
+    /* lane g: encode group g (deltas for uids[s+1+4g .. s+4+4g], zero-padded) */
+    u32 glen = 0;
+    u32 dl[4];
+    u8 ln[4];
+    if (active && lane < (int)ng) {
+        u64 prev = uids[s + (u64)(4 * lane)];
+        u32 tot = 1;
+        for (int k = 0; k < 4; k++) {
+            u64 idx = s + (u64)(4 * lane) + 1 + (u64)k;
+            u32 d = (idx < e) ? (u32)(uids[idx] - prev) : 0;
+            if (idx < e) prev = uids[idx];
+            dl[k] = d;
+            ln[k] = (u8)(1 + (d > 0xffu) + (d > 0xffffu) + (d > 0xffffffu));
+            tot += ln[k];
+        }
+        glen = tot;
+    }
+    /* wave exclusive scan of group byte-lengths */
+    u32 incl = glen;
+#pragma unroll
+    for (int o = 1; o < 64; o <<= 1) {
+        u32 x = __shfl_up(incl, o);
+        if (lane >= o) incl += x;
+    }
+    u32 goff = incl - glen;
+    u32 total = (u32)__shfl(incl, 63);
+    if (active && lane < (int)ng) {
+        u8 *p = &sbytes[wv][goff];
+        u8 tag = (u8)((ln[0] - 1) | ((ln[1] - 1) << 2) | ((ln[2] - 1) << 4) |
+                      ((ln[3] - 1) << 6));
+        *p++ = tag;
+        for (int k = 0; k < 4; k++) {
+            u32 d = dl[k];
+            for (int bcount = 0; bcount < ln[k]; bcount++) {
+                *p++ = (u8)(d & 0xff);
+                d >>= 8;
+            }
+        }
+    }
+    __syncthreads(); /* cross-lane LDS visibility before the copy-out */
+    if (!active) return;
+    u8 *gdst = stage + b * UA_ENC_STRIDE;
+    for (u32 i = lane; i < total; i += 64) gdst[i] = sbytes[wv][i];
+    if (lane == 0) blk_bytes[b] = total;
+}
+
+__global__ __launch_bounds__(UA_BLOCK) void k_enc_finalize(
+    const u8 *__restrict__ stage, const u32 *__restrict__ blk_bytes,
+    const u64 *__restrict__ offs, const u64 *__restrict__ partials, u64 nb,
+    u8 *__restrict__ deltas, u64 *__restrict__ delta_offs) {
+    u64 b = (u64)blockIdx.x * 4 + (threadIdx.x >> 6);
+    int lane = threadIdx.x & 63;
+    if (b > nb) return;
+    u64 off = d_off(offs, partials, b);
+    if (lane == 0) delta_offs[b] = off;
+    if (b == nb) return;
+    u32 bytes = blk_bytes[b];
+    const u8 *src = stage + b * UA_ENC_STRIDE;
+    u8 *dst = deltas + off;
+    for (u32 i = lane; i < bytes; i += 64) dst[i] = src[i];
+}
+
 /* ==================== host shim ==================== */
 
 static thread_local hipError_t g_last_hip = hipSuccess;
@@ -1172,6 +1300,83 @@ extern "C" int ua_decode_dev(ua_ctx *c, const ua_dpack *pk, uint64_t seek_uid, u
                              uint64_t *out_n) {
     std::lock_guard<std::mutex> g(c->mu);
     return run_packed_locked(c, pk, seek_uid, nullptr, 0, out, out_n, 1);
+}
+
+/* ---- GPU codec.Encode pipeline (codec.go:393 semantics, engine layout) ---- */
+extern "C" int ua_encode_dev(ua_ctx *c, const uint64_t *uids, uint64_t n,
+                             uint32_t block_size, uint64_t *bases, uint32_t *num_uids,
+                             uint64_t *delta_offs, uint8_t *deltas,
+                             uint64_t *n_blocks_out, uint64_t *deltas_bytes_out) {
+    std::lock_guard<std::mutex> g(c->mu);
+    HIP_TRY(hipSetDevice(c->device));
+    if (block_size > UA_MAX_BLOCK_UIDS) return UA_ERR_INVALID;
+    if (n == 0) {
+        HIP_TRY(hipMemsetAsync(delta_offs, 0, sizeof(u64), c->stream));
+        HIP_TRY(hipStreamSynchronize(c->stream));
+        *n_blocks_out = 0;
+        *deltas_bytes_out = 0;
+        return UA_OK;
+    }
+    int rc;
+    if ((rc = ws_reserve(c, WS_TCNT, (n + 1) * sizeof(u32)))) return rc;
+    if ((rc = ws_reserve(c, WS_TOFF, (n + 1) * sizeof(u64)))) return rc;
+    if ((rc = ws_reserve(c, WS_SCRATCH_A, (n + 1) * sizeof(u64)))) return rc;
+    if ((rc = ws_reserve(c, WS_SCRATCH_B, (n + 1) * sizeof(u64)))) return rc;
+    u32 *d_flags = (u32 *)c->ws[WS_TCNT];
+    u64 *d_offs = (u64 *)c->ws[WS_TOFF];
+    u64 *d_run = (u64 *)c->ws[WS_SCRATCH_A];
+    u64 *d_bstart = (u64 *)c->ws[WS_SCRATCH_B];
+
+    u64 nblk = (n + UA_BLOCK - 1) / UA_BLOCK;
+    u64 nblk1 = (n + 1 + UA_BLOCK - 1) / UA_BLOCK;
+    HIP_TRY(hipMemsetAsync(d_flags + n, 0, sizeof(u32), c->stream));
+    hipLaunchKernelGGL(k_enc_msb_flags, dim3((u32)nblk), dim3(UA_BLOCK), 0, c->stream,
+                       uids, n, d_flags);
+    if ((rc = run_scan(c, d_flags, n + 1, d_offs))) return rc;
+    u64 *d_part = (u64 *)c->ws[WS_PARTIAL];
+    hipLaunchKernelGGL(k_enc_mark_start, dim3((u32)nblk1), dim3(UA_BLOCK), 0, c->stream,
+                       d_flags, d_offs, d_part, n, d_run);
+    hipLaunchKernelGGL(k_enc_block_flags, dim3((u32)nblk), dim3(UA_BLOCK), 0, c->stream,
+                       d_flags, d_offs, d_part, d_run, n, block_size);
+    if ((rc = run_scan(c, d_flags, n + 1, d_offs))) return rc;
+    d_part = (u64 *)c->ws[WS_PARTIAL];
+    hipLaunchKernelGGL(k_enc_mark_start, dim3((u32)nblk1), dim3(UA_BLOCK), 0, c->stream,
+                       d_flags, d_offs, d_part, n, d_bstart);
+    u64 off_n = 0, part_n = 0;
+    HIP_TRY(hipMemcpyAsync(&off_n, d_offs + n, sizeof(u64), hipMemcpyDeviceToHost, c->stream));
+    HIP_TRY(hipMemcpyAsync(&part_n, d_part + n / UA_SCAN_CHUNK, sizeof(u64),
+                           hipMemcpyDeviceToHost, c->stream));
+    HIP_TRY(hipStreamSynchronize(c->stream));
+    u64 nb = off_n + part_n;
+
+    if ((rc = ws_reserve(c, WS_STAGE, (nb ? nb : 1) * UA_ENC_STRIDE))) return rc;
+    u8 *d_stage = (u8 *)c->ws[WS_STAGE];
+    HIP_TRY(hipEventRecord(c->ev[0], c->stream));
+    hipLaunchKernelGGL(k_encode_blocks, dim3((u32)((nb + UA_PKW - 1) / UA_PKW)),
+                       dim3(UA_BLOCK), 0, c->stream, uids, d_bstart, nb, bases, num_uids,
+                       d_flags, d_stage);
+    HIP_TRY(hipEventRecord(c->ev[1], c->stream));
+    HIP_TRY(hipMemsetAsync(d_flags + nb, 0, sizeof(u32), c->stream));
+    if ((rc = run_scan(c, d_flags, nb + 1, d_offs))) return rc;
+    d_part = (u64 *)c->ws[WS_PARTIAL];
+    hipLaunchKernelGGL(k_enc_finalize, dim3((u32)((nb + 1 + 3) / 4)), dim3(UA_BLOCK), 0,
+                       c->stream, d_stage, d_flags, d_offs, d_part, nb, deltas, delta_offs);
+    u64 db_off = 0, db_part = 0;
+    HIP_TRY(hipMemcpyAsync(&db_off, d_offs + nb, sizeof(u64), hipMemcpyDeviceToHost,
+                           c->stream));
+    HIP_TRY(hipMemcpyAsync(&db_part, d_part + nb / UA_SCAN_CHUNK, sizeof(u64),
+                           hipMemcpyDeviceToHost, c->stream));
+    HIP_TRY(hipStreamSynchronize(c->stream));
+    HIP_TRY(hipGetLastError());
+    *n_blocks_out = nb;
+    *deltas_bytes_out = db_off + db_part;
+
+    float ms = 0.f;
+    HIP_TRY(hipEventElapsedTime(&ms, c->ev[0], c->ev[1]));
+    c->kernel_ms += ms;
+    c->n_launches += 1;
+    c->bytes_algo += 8 * n + (db_off + db_part) + nb * 20;
+    return UA_OK;
 }
 
 /* ==================== host-side codec (codec.Encode restated) ==================== */

@@ -106,6 +106,17 @@ int ua_pack_flatten(const ua_pack *, uint64_t *bases, uint32_t *num_uids,
 int ua_decode_dev(ua_ctx *, const ua_dpack *, uint64_t seek_uid,
                   uint64_t *out, uint64_t *out_n);
 
+/* GPU codec.Encode (codec.go:393): uids (device, sorted) -> flattened pack
+ * arrays (device, caller-allocated worst case: bases/num_uids/delta_offs for
+ * max_blocks = n blocks (+1 for delta_offs), deltas capacity >= 6*n + 24
+ * bytes).  Byte-identical to the host encoder / reference format.
+ * block_size <= 256 (the engine's block bound); 0 packs per-uid blocks like
+ * the reference. */
+int ua_encode_dev(ua_ctx *, const uint64_t *uids, uint64_t n,
+                  uint32_t block_size, uint64_t *bases, uint32_t *num_uids,
+                  uint64_t *delta_offs, uint8_t *deltas,
+                  uint64_t *n_blocks_out, uint64_t *deltas_bytes_out);
+
 /* ---- batched device-resident set algebra ----
  * All pointers in ua_dpair are DEVICE pointers; the descriptor array itself
  * and out_lens live on the host.  Inputs are sorted uint64 lists; results are

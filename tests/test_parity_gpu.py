@@ -281,6 +281,58 @@ def test_packed_32msb_splits(eng):
     assert to_np(eng.decode_pack(dp, 0)).tolist() == uids.tolist()
 
 
+@pytest.mark.parametrize("size,bs", [(0, 256), (1, 256), (5, 0), (300, 10),
+                                     (5000, 1), (200_000, 256), (2_000_000, 256)])
+def test_encode_dev_matches_oracle_bytes(eng, size, bs):
+    """GPU codec.Encode vs the oracle encoder: byte-identical flat pack
+    (blocks split on 32-MSB + block_size, group-varint deltas)."""
+    rng = np.random.default_rng(SEED + size + bs)
+    uids = synth.getuids_geometric(rng, max(size, 1))[:size]
+    if size > 10:  # force 32-MSB splits (codec.go:117)
+        uids = uids.copy()
+        uids[size // 2:] += np.uint64(1) << np.uint64(33)
+        uids[3 * size // 4:] += np.uint64(1) << np.uint64(41)
+    dp = eng.encode_dev(to_dev(uids), bs)
+
+    opack = orc.Pack(uids, bs)
+    obases, onums, ooffs, oblob = opack.flatten()
+    assert dp.bases.numel() == obases.size
+    assert to_np(dp.bases).tolist() == obases.tolist()
+    assert dp.num_uids.cpu().numpy().view(np.uint32).tolist() == onums.tolist()
+    assert to_np(dp.delta_offs).tolist() == ooffs.tolist()
+    got_blob = dp.deltas.cpu().numpy().view(np.uint8)[:oblob.size]
+    assert got_blob.tolist() == oblob.tolist()
+    if size:
+        # and the GPU decoder inverts the GPU encoder
+        assert to_np(eng.decode_pack(dp)).tolist() == uids.tolist()
+
+
+def test_packed_compositions_vs_oracle(eng):
+    """MergeSortedPacked / IntersectSortedPacked / ApplyFilterPacked
+    (algo/packed.go:222,100,16) as engine compositions."""
+    rng = np.random.default_rng(SEED)
+    lists = [synth.gen_sorted_unique(rng, int(rng.integers(100, 20_000)), 60_000)
+             for _ in range(5)]
+    d_lists = [to_dev(x) for x in lists]
+
+    mp = eng.merge_sorted_packed(d_lists, 256)
+    want_merge = orc.merge_sorted(lists)
+    assert to_np(eng.decode_pack(mp)).tolist() == want_merge.tolist()
+    # byte parity of the resulting pack vs oracle-encoding the oracle merge
+    ob, on, oo, obl = orc.Pack(want_merge, 256).flatten()
+    assert to_np(mp.bases).tolist() == ob.tolist()
+    assert mp.deltas.cpu().numpy().view(np.uint8)[:obl.size].tolist() == obl.tolist()
+
+    dpacks = [eng.encode_dev(t, 10) for t in d_lists]
+    ip = eng.intersect_sorted_packed(dpacks)
+    want_i = orc.intersect_sorted(lists)
+    assert to_np(eng.decode_pack(ip)).tolist() == want_i.tolist()
+
+    fp = eng.apply_filter_packed(dpacks[0], lambda t: (t % 2) == 1)
+    want_f = lists[0][lists[0] % 2 == 1]
+    assert to_np(eng.decode_pack(fp)).tolist() == want_f.tolist()
+
+
 def test_host_packed_api(eng):
     """ua_intersect_packed host-pointer convenience (the cgo surface)."""
     import ctypes as Ct

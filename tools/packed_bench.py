@@ -63,8 +63,23 @@ def main():
     el_d = time.perf_counter() - t0
     st_d = eng.stats()
 
+    # GPU encode (ua_encode_dev)
+    d_uids = torch.from_numpy(pack_uids.view(np.int64)).cuda()
+    dp2 = eng.encode_dev(d_uids, 256)  # warm
+    assert dp2.bases.numel() == bases.size
+    torch.cuda.synchronize()
+    eng.stats_reset()
+    t0 = time.perf_counter()
+    for _ in range(steps):
+        eng.encode_dev(d_uids, 256)
+    torch.cuda.synchronize()
+    el_e = time.perf_counter() - t0
+    st_e = eng.stats()
+
     res = {
         "workload": "cfg4_packed_21M",
+        "gpu_encode_Muids_per_s": round(pack_uids.size * steps / el_e / 1e6, 1),
+        "gpu_encode_kernel_ms": round(st_e["kernel_ms"] / max(st_e["launches"], 1), 4),
         "pack_uids": int(pack_uids.size),
         "pack_bytes": int(blob.size + 20 * bases.size),
         "bytes_per_uid": round((blob.size + 20 * bases.size) / pack_uids.size, 3),
