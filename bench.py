@@ -50,7 +50,7 @@ def load_traffic(pairs):
     if os.path.exists(path):
         try:
             d = json.load(open(path))
-            if f"{pairs} pairs/launch" in d.get("workload", ""):
+            if d.get("pairs_per_launch") == pairs:
                 return d.get("bytes_per_launch")
         except Exception:
             return None
@@ -95,7 +95,8 @@ def main():
     world = int(os.environ.get("WORLD_SIZE", "1"))
     rank = int(os.environ.get("RANK", "0"))
     local_rank = int(os.environ.get("LOCAL_RANK", "0"))
-    distributed = world > 1
+    # UA_BENCH_FORCE_DIST exercises the torch.distributed path at world==1
+    distributed = world > 1 or bool(os.environ.get("UA_BENCH_FORCE_DIST"))
     if distributed:
         import torch.distributed as dist
         backend = "nccl" if torch.cuda.is_available() else "gloo"

@@ -871,18 +871,18 @@ extern "C" int ua_ctx_create(ua_ctx **out, int device) {
 
 extern "C" void ua_ctx_destroy(ua_ctx *c) {
     if (!c) return;
-    hipSetDevice(c->device);
+    (void)hipSetDevice(c->device);
     for (int i = 0; i < WS_COUNT; i++)
-        if (c->ws[i]) hipFree(c->ws[i]);
+        if (c->ws[i]) (void)hipFree(c->ws[i]);
     for (int i = 0; i < 4; i++)
-        if (c->ev[i]) hipEventDestroy(c->ev[i]);
-    if (c->stream) hipStreamDestroy(c->stream);
+        if (c->ev[i]) (void)hipEventDestroy(c->ev[i]);
+    if (c->stream) (void)hipStreamDestroy(c->stream);
     delete c;
 }
 
 static int ws_reserve(ua_ctx *c, int slot, size_t bytes) {
     if (bytes <= c->ws_cap[slot]) return UA_OK;
-    if (c->ws[slot]) hipFree(c->ws[slot]);
+    if (c->ws[slot]) (void)hipFree(c->ws[slot]);
     c->ws[slot] = nullptr;
     c->ws_cap[slot] = 0;
     size_t cap = bytes + bytes / 4; /* 25% headroom limits realloc churn */

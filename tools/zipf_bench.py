@@ -1,0 +1,68 @@
+#!/usr/bin/env python3
+"""cfg 3 shape: one grid over thousands of Zipf-skewed
+predicate list-pairs — intersect, pairwise MergeSorted, Difference.
+Validates the batched launcher under size skew and reports throughput.
+Prints one JSON line; parity for this shape is gated by
+tests/test_parity_gpu.py::test_batched_zipf_vs_oracle.
+"""
+import json
+import os
+import sys
+import time
+
+import numpy as np
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+import torch  # noqa: E402
+
+from dgraph_amd import algo, synth  # noqa: E402
+
+
+def gen_sorted(rng, n):
+    deltas = rng.integers(1, 17, size=n, dtype=np.uint64)
+    return (np.uint64(rng.integers(0, 1000)) + np.cumsum(deltas)).astype(np.uint64)
+
+
+def main():
+    pairs = int(sys.argv[1]) if len(sys.argv) > 1 else 4096
+    steps = int(sys.argv[2]) if len(sys.argv) > 2 else 20
+    rng = np.random.default_rng(synth.SEED + 3)
+    sizes_u = synth.zipf_sizes(rng, pairs, lo=1000, hi=2_000_000)
+    sizes_v = synth.zipf_sizes(rng, pairs, lo=1000, hi=2_000_000)
+    total = int(sizes_u.sum() + sizes_v.sum())
+    eng = algo.Engine(0)
+    us, vs = [], []
+    for p in range(pairs):
+        us.append(torch.from_numpy(gen_sorted(rng, int(sizes_u[p])).view(np.int64)).cuda())
+        vs.append(torch.from_numpy(gen_sorted(rng, int(sizes_v[p])).view(np.int64)).cuda())
+    i_outs = [torch.empty(min(u.numel(), v.numel()), dtype=torch.int64, device="cuda")
+              for u, v in zip(us, vs)]
+    m_outs = [torch.empty(u.numel() + v.numel(), dtype=torch.int64, device="cuda")
+              for u, v in zip(us, vs)]
+    d_outs = [torch.empty(max(u.numel(), 1), dtype=torch.int64, device="cuda") for u in us]
+
+    res = {"workload": "cfg3_zipf_batch", "pairs": pairs,
+           "total_elems": total, "total_MB": round(total * 8 / 1e6, 1),
+           "size_min": int(min(sizes_u.min(), sizes_v.min())),
+           "size_max": int(max(sizes_u.max(), sizes_v.max()))}
+    for name, fn, outs in [("intersect", eng.intersect_pairs, i_outs),
+                           ("merge", eng.merge_pairs, m_outs),
+                           ("difference", eng.difference_pairs, d_outs)]:
+        fn(us, vs, outs)
+        torch.cuda.synchronize()
+        eng.stats_reset()
+        t0 = time.perf_counter()
+        for _ in range(steps):
+            fn(us, vs, outs)
+        torch.cuda.synchronize()
+        el = time.perf_counter() - t0
+        st = eng.stats()
+        res[f"{name}_ms_per_batch"] = round(el / steps * 1e3, 3)
+        res[f"{name}_Gelems_per_s"] = round(total * steps / el / 1e9, 1)
+        res[f"{name}_kernel_ms"] = round(st["kernel_ms"] / max(st["launches"], 1), 3)
+    print(json.dumps(res))
+
+
+if __name__ == "__main__":
+    main()
