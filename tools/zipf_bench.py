@@ -28,8 +28,8 @@ def main():
     pairs = int(sys.argv[1]) if len(sys.argv) > 1 else 4096
     steps = int(sys.argv[2]) if len(sys.argv) > 2 else 20
     rng = np.random.default_rng(synth.SEED + 3)
-    sizes_u = synth.zipf_sizes(rng, pairs, lo=1000, hi=2_000_000)
-    sizes_v = synth.zipf_sizes(rng, pairs, lo=1000, hi=2_000_000)
+    sizes_u = synth.zipf_sizes(rng, pairs, lo=1000, hi=500_000)
+    sizes_v = synth.zipf_sizes(rng, pairs, lo=1000, hi=500_000)
     total = int(sizes_u.sum() + sizes_v.sum())
     eng = algo.Engine(0)
     us, vs = [], []
