@@ -84,6 +84,10 @@ def lib():
         L.ua_stats_reset.argtypes = [C.c_void_p]
         L.ua_stats_get.argtypes = [C.c_void_p, _u64p, C.POINTER(C.c_double), _u64p]
         L.ua_intersect_batch_dev.argtypes = [C.c_void_p, C.POINTER(UaDPair), C.c_int, _u64p]
+        L.ua_batch_create.argtypes = [C.c_void_p, C.POINTER(UaDPair), C.c_int, _voidpp]
+        L.ua_batch_run.argtypes = [C.c_void_p, C.c_void_p, C.c_int, _u64p]
+        L.ua_batch_destroy.argtypes = [C.c_void_p, C.c_void_p]
+        L.ua_batch_destroy.restype = None
         L.ua_merge_batch_dev.argtypes = [C.c_void_p, C.POINTER(UaDPair), C.c_int, _u64p]
         L.ua_difference_batch_dev.argtypes = [C.c_void_p, C.POINTER(UaDPair), C.c_int, _u64p]
         L.ua_index_of_batch_dev.argtypes = [C.c_void_p, C.c_void_p, _u64, C.c_void_p, _u64,

@@ -109,6 +109,22 @@ class Engine:
         lens = self._run_pairs(lib().ua_difference_batch_dev, us, vs, outs)
         return outs, lens
 
+    def make_batch(self, us, vs, outs):
+        """Prepared batch: descriptor upload + merge-path partition done once
+        (ua_batch_create); run ops repeatedly with Batch.run().  The input
+        tensors' contents must stay unchanged while the batch lives."""
+        np_ = len(us)
+        pairs = (UaDPair * np_)()
+        for i in range(np_):
+            pairs[i].u = us[i].data_ptr()
+            pairs[i].n = us[i].numel()
+            pairs[i].v = vs[i].data_ptr()
+            pairs[i].m = vs[i].numel()
+            pairs[i].out = outs[i].data_ptr()
+        h = C.c_void_p()
+        check(lib().ua_batch_create(self._ctx, pairs, np_, C.byref(h)))
+        return Batch(self, h, np_, (us, vs, outs))
+
     def intersect_sorted(self, lists):
         """algo.IntersectSorted (uidlist.go:297): k-way fold, smallest first."""
         import torch
@@ -234,6 +250,38 @@ class Engine:
         check(lib().ua_decode_dev(self._ctx, C.byref(pk), _u64(seek),
                                   C.c_void_p(out.data_ptr()), C.byref(out_n)))
         return out[:out_n.value]
+
+
+OP_INTERSECT = 0
+OP_MERGE = 1
+OP_DIFFERENCE = 2
+
+
+class Batch:
+    """Prepared pair batch (ua_batch): run() executes one op over all pairs
+    as one grid, returning per-pair output lengths."""
+
+    def __init__(self, engine, handle, n_pairs, keepalive):
+        self._eng = engine
+        self._h = handle
+        self.n_pairs = n_pairs
+        self._keep = keepalive  # input/output tensors must outlive the batch
+
+    def run(self, op=OP_INTERSECT):
+        lens = (C.c_uint64 * self.n_pairs)()
+        check(lib().ua_batch_run(self._eng._ctx, self._h, op, lens))
+        return [int(lens[i]) for i in range(self.n_pairs)]
+
+    def close(self):
+        if self._h:
+            lib().ua_batch_destroy(self._eng._ctx, self._h)
+            self._h = C.c_void_p()
+
+    def __del__(self):
+        try:
+            self.close()
+        except Exception:
+            pass
 
 
 class DPack:

@@ -1103,6 +1103,187 @@ extern "C" int ua_intersect_batch_dev(ua_ctx *c, const ua_dpair *pairs, int n_pa
     return run_batch(c, pairs, n_pairs, out_lens, OP_INTERSECT);
 }
 
+/* ==================== prepared batch (repeated-query path) ==================== */
+
+struct ua_batch {
+    int n_pairs = 0;
+    u64 total_tiles = 0;
+    u64 in_bytes = 0;
+    u64 nchunks = 0;
+    void *mem = nullptr;     /* one allocation for all metadata arrays */
+    u64 *d_stage = nullptr;  /* lazy; stride UA_TILE (fits intersect + diff) */
+    UaDesc *d_descs = nullptr;
+    u64 *d_tb = nullptr;
+    u32 *d_tpair = nullptr;
+    u32 *d_ta0 = nullptr;
+    u32 *d_tcnt = nullptr;
+    u64 *d_toff = nullptr;
+    u64 *d_part = nullptr;
+    u64 *d_pout = nullptr;
+};
+
+static size_t align16(size_t x) { return (x + 15) & ~(size_t)15; }
+
+extern "C" int ua_batch_create(ua_ctx *c, const ua_dpair *pairs, int n_pairs,
+                               ua_batch **out) {
+    std::lock_guard<std::mutex> g(c->mu);
+    HIP_TRY(hipSetDevice(c->device));
+    ua_batch *b = new ua_batch();
+    b->n_pairs = n_pairs;
+    std::vector<UaDesc> descs((size_t)std::max(n_pairs, 1));
+    std::vector<u64> tb((size_t)n_pairs + 1);
+    for (int p = 0; p < n_pairs; p++) {
+        const ua_dpair &pr = pairs[p];
+        if (pr.n >= (1ull << 31) || pr.m >= (1ull << 31)) {
+            delete b;
+            return UA_ERR_INVALID;
+        }
+        descs[p] = {pr.u, pr.n, pr.v, pr.m, pr.out, b->total_tiles};
+        tb[p] = b->total_tiles;
+        b->total_tiles += (pr.n + pr.m + UA_TILE - 1) / UA_TILE;
+        b->in_bytes += 8 * (pr.n + pr.m);
+    }
+    tb[n_pairs] = b->total_tiles;
+    u64 T = b->total_tiles;
+    b->nchunks = (T + 1 + UA_SCAN_CHUNK - 1) / UA_SCAN_CHUNK;
+
+    size_t o_desc = 0;
+    size_t o_tb = align16(o_desc + descs.size() * sizeof(UaDesc));
+    size_t o_toff = align16(o_tb + tb.size() * sizeof(u64));
+    size_t o_part = align16(o_toff + (T + 1) * sizeof(u64));
+    size_t o_pout = align16(o_part + (b->nchunks + 1) * sizeof(u64));
+    size_t o_tpair = align16(o_pout + (size_t)std::max(n_pairs, 1) * sizeof(u64));
+    size_t o_ta0 = align16(o_tpair + (T + 1) * sizeof(u32));
+    size_t o_tcnt = align16(o_ta0 + (T + 1) * sizeof(u32));
+    size_t total_bytes = align16(o_tcnt + (T + 1) * sizeof(u32));
+    hipError_t e = hipMalloc(&b->mem, total_bytes);
+    if (e != hipSuccess) {
+        g_last_hip = e;
+        delete b;
+        return UA_ERR_NOMEM;
+    }
+    u8 *base = (u8 *)b->mem;
+    b->d_descs = (UaDesc *)(base + o_desc);
+    b->d_tb = (u64 *)(base + o_tb);
+    b->d_toff = (u64 *)(base + o_toff);
+    b->d_part = (u64 *)(base + o_part);
+    b->d_pout = (u64 *)(base + o_pout);
+    b->d_tpair = (u32 *)(base + o_tpair);
+    b->d_ta0 = (u32 *)(base + o_ta0);
+    b->d_tcnt = (u32 *)(base + o_tcnt);
+
+    std::vector<u8> hostbuf(o_tb + tb.size() * sizeof(u64));
+    memcpy(hostbuf.data(), descs.data(), descs.size() * sizeof(UaDesc));
+    memcpy(hostbuf.data() + o_tb, tb.data(), tb.size() * sizeof(u64));
+    HIP_TRY(hipMemcpyAsync(b->mem, hostbuf.data(), hostbuf.size(),
+                           hipMemcpyHostToDevice, c->stream));
+    HIP_TRY(hipMemsetAsync(b->d_tcnt + T, 0, sizeof(u32), c->stream));
+    HIP_TRY(hipMemsetAsync(b->d_toff, 0, sizeof(u64), c->stream));
+    HIP_TRY(hipMemsetAsync(b->d_part, 0, sizeof(u64), c->stream));
+    if (T > 0) {
+        /* the partition depends only on the (immutable) pair contents:
+         * computed once here, reused every run */
+        u64 pblk = (T + UA_BLOCK - 1) / UA_BLOCK;
+        hipLaunchKernelGGL(k_partition, dim3((u32)pblk), dim3(UA_BLOCK), 0, c->stream,
+                           b->d_descs, b->d_tb, n_pairs, T, b->d_tpair, b->d_ta0, 0);
+        hipLaunchKernelGGL(k_partition, dim3((u32)pblk), dim3(UA_BLOCK), 0, c->stream,
+                           b->d_descs, b->d_tb, n_pairs, T, b->d_tpair, b->d_ta0, 1);
+    }
+    HIP_TRY(hipStreamSynchronize(c->stream));
+    HIP_TRY(hipGetLastError());
+    *out = b;
+    return UA_OK;
+}
+
+extern "C" void ua_batch_destroy(ua_ctx *c, ua_batch *b) {
+    if (!b) return;
+    (void)hipSetDevice(c->device);
+    if (b->mem) (void)hipFree(b->mem);
+    if (b->d_stage) (void)hipFree(b->d_stage);
+    delete b;
+}
+
+extern "C" int ua_batch_run(ua_ctx *c, ua_batch *b, int op, uint64_t *out_lens) {
+    std::lock_guard<std::mutex> g(c->mu);
+    HIP_TRY(hipSetDevice(c->device));
+    if (b->n_pairs == 0) return UA_OK;
+    int kop = (op == UA_OP_INTERSECT) ? OP_INTERSECT
+              : (op == UA_OP_MERGE) ? OP_UNION
+              : (op == UA_OP_DIFFERENCE) ? OP_DIFF
+                                         : -1;
+    if (kop < 0) return UA_ERR_INVALID;
+    u64 T = b->total_tiles;
+    u64 stride = (kop == OP_INTERSECT) ? UA_TILE / 2 : UA_TILE;
+    if (kop != OP_UNION && !b->d_stage) {
+        hipError_t e = hipMalloc((void **)&b->d_stage, (T ? T : 1) * UA_TILE * sizeof(u64));
+        if (e != hipSuccess) {
+            g_last_hip = e;
+            return UA_ERR_NOMEM;
+        }
+    }
+    if (T > 0) {
+        HIP_TRY(hipEventRecord(c->ev[0], c->stream));
+        if (kop == OP_INTERSECT) {
+            hipLaunchKernelGGL((k_tiles<OP_INTERSECT, MODE_STAGE>), dim3((u32)T),
+                               dim3(UA_BLOCK), 0, c->stream, b->d_descs, b->d_tpair,
+                               b->d_ta0, T, b->d_stage, stride, b->d_tcnt,
+                               (u64 *)nullptr, (u64 *)nullptr);
+        } else if (kop == OP_DIFF) {
+            hipLaunchKernelGGL((k_tiles<OP_DIFF, MODE_STAGE>), dim3((u32)T),
+                               dim3(UA_BLOCK), 0, c->stream, b->d_descs, b->d_tpair,
+                               b->d_ta0, T, b->d_stage, stride, b->d_tcnt,
+                               (u64 *)nullptr, (u64 *)nullptr);
+        } else {
+            hipLaunchKernelGGL((k_tiles<OP_UNION, MODE_COUNT>), dim3((u32)T),
+                               dim3(UA_BLOCK), 0, c->stream, b->d_descs, b->d_tpair,
+                               b->d_ta0, T, (u64 *)nullptr, 0, b->d_tcnt,
+                               (u64 *)nullptr, (u64 *)nullptr);
+        }
+        HIP_TRY(hipEventRecord(c->ev[1], c->stream));
+
+        hipLaunchKernelGGL(k_scan1, dim3((u32)b->nchunks), dim3(UA_BLOCK), 0, c->stream,
+                           b->d_tcnt, T + 1, b->d_toff, b->d_part);
+        hipLaunchKernelGGL(k_scan2, dim3(1), dim3(UA_BLOCK), 0, c->stream, b->d_part,
+                           b->nchunks);
+        if (kop == OP_UNION) {
+            HIP_TRY(hipEventRecord(c->ev[2], c->stream));
+            hipLaunchKernelGGL((k_tiles<OP_UNION, MODE_WRITE>), dim3((u32)T),
+                               dim3(UA_BLOCK), 0, c->stream, b->d_descs, b->d_tpair,
+                               b->d_ta0, T, (u64 *)nullptr, 0, b->d_tcnt, b->d_toff,
+                               b->d_part);
+            HIP_TRY(hipEventRecord(c->ev[3], c->stream));
+        } else {
+            hipLaunchKernelGGL(k_compact, dim3((u32)((T + 15) / 16)), dim3(UA_BLOCK), 0,
+                               c->stream, b->d_descs, b->d_tpair, b->d_tcnt, b->d_toff,
+                               b->d_part, b->d_stage, stride, T);
+        }
+    }
+    u64 poutblk = ((u64)b->n_pairs + UA_BLOCK - 1) / UA_BLOCK;
+    hipLaunchKernelGGL(k_pair_out, dim3((u32)poutblk), dim3(UA_BLOCK), 0, c->stream,
+                       b->d_toff, b->d_part, b->d_tb, b->n_pairs, b->d_pout);
+    HIP_TRY(hipMemcpyAsync(out_lens, b->d_pout, (size_t)b->n_pairs * sizeof(u64),
+                           hipMemcpyDeviceToHost, c->stream));
+    HIP_TRY(hipStreamSynchronize(c->stream));
+    HIP_TRY(hipGetLastError());
+
+    if (T > 0) {
+        float ms = 0.f;
+        HIP_TRY(hipEventElapsedTime(&ms, c->ev[0], c->ev[1]));
+        c->kernel_ms += ms;
+        c->n_launches += 1;
+        if (kop == OP_UNION) {
+            float ms2 = 0.f;
+            HIP_TRY(hipEventElapsedTime(&ms2, c->ev[2], c->ev[3]));
+            c->kernel_ms += ms2;
+            c->n_launches += 1;
+        }
+    }
+    u64 out_elems = 0;
+    for (int p = 0; p < b->n_pairs; p++) out_elems += out_lens[p];
+    c->bytes_algo += b->in_bytes + 8 * out_elems;
+    return UA_OK;
+}
+
 extern "C" int ua_merge_batch_dev(ua_ctx *c, const ua_dpair *pairs, int n_pairs,
                                   uint64_t *out_lens) {
     return run_batch(c, pairs, n_pairs, out_lens, OP_UNION);

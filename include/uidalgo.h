@@ -134,6 +134,17 @@ typedef struct {
 /* batched algo.IntersectWith (uidlist.go:142) */
 int ua_intersect_batch_dev(ua_ctx *, const ua_dpair *pairs, int n_pairs,
                            uint64_t *out_lens);
+
+/* Prepared batch (the repeated-query path): descriptor upload and the
+ * merge-path tile partition are done ONCE at create; each run is launches
+ * only.  The pairs' device contents must not change between create and the
+ * runs (the cached partition depends on them); out capacities must fit the
+ * op (intersect >= min(n,m), merge >= n+m, difference >= n). */
+typedef struct ua_batch ua_batch;
+enum { UA_OP_INTERSECT = 0, UA_OP_MERGE = 1, UA_OP_DIFFERENCE = 2 };
+int ua_batch_create(ua_ctx *, const ua_dpair *pairs, int n_pairs, ua_batch **out);
+int ua_batch_run(ua_ctx *, ua_batch *, int op, uint64_t *out_lens);
+void ua_batch_destroy(ua_ctx *, ua_batch *);
 /* batched pairwise algo.MergeSorted semantics (dedup union, uidlist.go:448) */
 int ua_merge_batch_dev(ua_ctx *, const ua_dpair *pairs, int n_pairs,
                        uint64_t *out_lens);

@@ -186,6 +186,34 @@ def test_batched_zipf_vs_oracle(eng):
             orc.difference(us[i], vs[i]).tolist(), f"diff pair {i}"
 
 
+def test_prepared_batch_vs_oracle(eng):
+    """ua_batch: partition cached at create; repeated runs of all three ops
+    over one prepared batch are bit-exact and stable across runs."""
+    import torch
+    rng = np.random.default_rng(SEED + 77)
+    us, vs = [], []
+    for _ in range(32):
+        n = int(rng.integers(0, 50_000))
+        m = int(rng.integers(0, 50_000))
+        us.append(synth.gen_sorted_unique(rng, n, 3 * (n + m) + 10))
+        vs.append(synth.gen_sorted_unique(rng, m, 3 * (n + m) + 10))
+    d_us = [to_dev(x) for x in us]
+    d_vs = [to_dev(x) for x in vs]
+    # out capacity n+m fits every op
+    outs = [torch.empty(max(u.numel() + v.numel(), 1), dtype=torch.int64,
+                        device="cuda:0") for u, v in zip(d_us, d_vs)]
+    batch = eng.make_batch(d_us, d_vs, outs)
+    for op, ref in [(algo.OP_INTERSECT, orc.intersect_with),
+                    (algo.OP_DIFFERENCE, orc.difference),
+                    (algo.OP_MERGE, lambda u, v: orc.merge_sorted([u, v])),
+                    (algo.OP_INTERSECT, orc.intersect_with)]:  # re-run intersect
+        lens = batch.run(op)
+        for i in range(len(us)):
+            got = to_np(outs[i][:lens[i]])
+            assert got.tolist() == ref(us[i], vs[i]).tolist(), f"op={op} pair={i}"
+    batch.close()
+
+
 @pytest.mark.parametrize("k", [1, 2, 3, 8, 150])
 def test_kway_vs_oracle(eng, k):
     rng = np.random.default_rng(SEED + k)
