@@ -312,6 +312,73 @@ __device__ __forceinline__ int tile_search(const u64 *As, int alen, const u64 *B
     return cnt;
 }
 
+/* Branchless walk: the asm of the if/else walk is ~25% exec-mask control
+ * (s_and_saveexec/s_or_b64/s_cbranch per step).  This variant keeps a
+ * 2-deep frontier per side in registers and does ONE unconditional LDS
+ * gather per step at a selected clamped address — no divergent branches in
+ * the loop body.  As and Bs must live in the SAME LDS array (smem) so the
+ * refill address can select between them.  aoff/boff are u64-element
+ * offsets of As/Bs within smem. */
+template <int OP>
+__device__ __forceinline__ int tile_walk2(const u64 *smembase, int aoff, int alen,
+                                          int boff, int blen, u64 a_before, bool has_ab,
+                                          bool has_bn, int s0, int s1, int i0,
+                                          u64 (&em)[UA_WPT], u32 &flags) {
+    const u64 *As = smembase + aoff;
+    const u64 *Bs = smembase + boff;
+    int i = i0, j = s0 - i0;
+    int cnt = 0;
+    flags = 0;
+    int steps = s1 - s0;
+    int blen_ext = blen + (has_bn ? 1 : 0);
+    int amax = alen > 0 ? alen - 1 : 0;
+    int bmax = blen_ext > 0 ? blen_ext - 1 : 0;
+    /* frontier: cur + next per side, clamped loads (garbage guarded by
+     * i/j bound checks in the predicates) */
+    u64 a = As[i < alen ? i : amax];
+    u64 an = As[(i + 1) < alen ? (i + 1) : amax];
+    u64 b = Bs[j < blen_ext ? j : bmax];
+    u64 bn = Bs[(j + 1) < blen_ext ? (j + 1) : bmax];
+    u64 prev_a = (i > 0) ? As[i - 1] : a_before;
+#pragma unroll
+    for (int s = 0; s < UA_WPT; s++) {
+        if (s >= steps) break;
+        bool inA = i < alen, inB = j < blen;
+        if (!inA && !inB) break;
+        bool takeA = inA && (!inB || a <= b);
+        bool eq = (a == b) && (j < blen_ext);
+        bool emit;
+        if (OP == OP_INTERSECT) {
+            emit = takeA && eq;
+            em[s] = a;
+        } else if (OP == OP_DIFF) {
+            emit = takeA && !eq;
+            em[s] = a;
+        } else { /* UNION */
+            bool dupB = (i > 0 || has_ab) && (b == prev_a);
+            emit = takeA || !dupB;
+            em[s] = takeA ? a : b;
+            prev_a = takeA ? a : prev_a;
+        }
+        flags |= ((u32)emit) << s;
+        cnt += emit;
+        int ni = i + (takeA ? 1 : 0);
+        int nj = j + (takeA ? 0 : 1);
+        /* one refill: next lookahead of the consumed side */
+        int ra = (ni + 1) < alen ? (ni + 1) : amax;
+        int rb = (nj + 1) < blen_ext ? (nj + 1) : bmax;
+        int raddr = takeA ? (aoff + ra) : (boff + rb);
+        u64 r = smembase[raddr];
+        a = takeA ? an : a;
+        b = takeA ? b : bn;
+        an = takeA ? r : an;
+        bn = takeA ? bn : r;
+        i = ni;
+        j = nj;
+    }
+    return cnt;
+}
+
 /* cooperative global->LDS fill, 16-B vectorized on the aligned body
  * (8-B/lane loads cap ~60% of the dwordx4 HBM rate — guide §2/G13) */
 __device__ __forceinline__ void d_fill_lds(u64 *dst, const u64 *__restrict__ src,
@@ -387,6 +454,9 @@ __global__ __launch_bounds__(UA_BLOCK) void k_tiles(
     u64 em[UA_WPT];
     u32 flags;
     int cnt;
+#ifndef UA_WALK2
+#define UA_WALK2 1 /* 0 = the branchy register-frontier walk */
+#endif
     if (UA_SEARCH && OP != OP_UNION) {
         cnt = tile_search<OP>(As, alen, Bs, blen, has_bn, tid, em, flags);
     } else {
@@ -397,8 +467,13 @@ __global__ __launch_bounds__(UA_BLOCK) void k_tiles(
         if (s1 > tilelen) s1 = tilelen;
         int i0 = d_merge_path_lds(As, alen, Bs, blen, s0);
         u64 a_before = s_abefore;
-        cnt = tile_walk<OP>(As, alen, Bs, blen, a_before, has_ab, has_bn, s0, s1, i0,
-                            em, flags);
+        if (UA_WALK2) {
+            cnt = tile_walk2<OP>(smem, 0, alen, (int)(Bs - smem), blen, a_before,
+                                 has_ab, has_bn, s0, s1, i0, em, flags);
+        } else {
+            cnt = tile_walk<OP>(As, alen, Bs, blen, a_before, has_ab, has_bn, s0, s1,
+                                i0, em, flags);
+        }
     }
 #endif
 #if UA_ABLATE == 3 /* fill+walk, skip scan/write-back */

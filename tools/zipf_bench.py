@@ -36,25 +36,23 @@ def main():
     for p in range(pairs):
         us.append(torch.from_numpy(gen_sorted(rng, int(sizes_u[p])).view(np.int64)).cuda())
         vs.append(torch.from_numpy(gen_sorted(rng, int(sizes_v[p])).view(np.int64)).cuda())
-    i_outs = [torch.empty(min(u.numel(), v.numel()), dtype=torch.int64, device="cuda")
-              for u, v in zip(us, vs)]
     m_outs = [torch.empty(u.numel() + v.numel(), dtype=torch.int64, device="cuda")
               for u, v in zip(us, vs)]
-    d_outs = [torch.empty(max(u.numel(), 1), dtype=torch.int64, device="cuda") for u in us]
 
     res = {"workload": "cfg3_zipf_batch", "pairs": pairs,
            "total_elems": total, "total_MB": round(total * 8 / 1e6, 1),
            "size_min": int(min(sizes_u.min(), sizes_v.min())),
            "size_max": int(max(sizes_u.max(), sizes_v.max()))}
-    for name, fn, outs in [("intersect", eng.intersect_pairs, i_outs),
-                           ("merge", eng.merge_pairs, m_outs),
-                           ("difference", eng.difference_pairs, d_outs)]:
-        fn(us, vs, outs)
+    # prepared batch with n+m capacity fits all three ops
+    batch = eng.make_batch(us, vs, m_outs)
+    for name, op in [("intersect", algo.OP_INTERSECT), ("merge", algo.OP_MERGE),
+                     ("difference", algo.OP_DIFFERENCE)]:
+        batch.run(op)
         torch.cuda.synchronize()
         eng.stats_reset()
         t0 = time.perf_counter()
         for _ in range(steps):
-            fn(us, vs, outs)
+            batch.run(op)
         torch.cuda.synchronize()
         el = time.perf_counter() - t0
         st = eng.stats()
