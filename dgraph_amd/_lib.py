@@ -42,6 +42,15 @@ class UaDPack(C.Structure):
     ]
 
 
+class UaPTask(C.Structure):
+    _fields_ = [
+        ("v", C.c_void_p),
+        ("m", _u64),
+        ("out", C.c_void_p),
+        ("after_uid", _u64),
+    ]
+
+
 class UaBlock(C.Structure):
     _fields_ = [
         ("base", _u64),
@@ -96,6 +105,9 @@ def lib():
         L.ua_merge_k_dev.argtypes = [C.c_void_p, _voidpp, _u64p, C.c_int, C.c_void_p, _u64p]
         L.ua_intersect_packed_dev.argtypes = [C.c_void_p, C.POINTER(UaDPack), _u64, C.c_void_p,
                                               _u64, C.c_void_p, _u64p]
+        L.ua_intersect_packed_batch_dev.argtypes = [
+            C.c_void_p, C.c_void_p, C.c_void_p, C.c_void_p, C.c_void_p,
+            _u64p, C.c_int, C.POINTER(UaPTask), _u64p]
         L.ua_decode_dev.argtypes = [C.c_void_p, C.POINTER(UaDPack), _u64, C.c_void_p, _u64p]
         L.ua_encode.argtypes = [_u64p, _u64, C.c_uint32, _voidpp]
         L.ua_encode_dev.argtypes = [C.c_void_p, C.c_void_p, _u64, C.c_uint32,

@@ -182,6 +182,63 @@ class Engine:
         total = int(np.asarray(num_uids, dtype=np.uint64).sum())
         return DPack(t_bases, t_nums, t_offs, t_blob, int(block_size), total)
 
+    def upload_pack_batch(self, packs):
+        """Concatenate many flat packs (tuples from encode_flat) into one
+        device arena + pack boundaries — the collected form of a
+        handleUidPostings fan-out (worker/task.go:834-971)."""
+        import torch
+        dev = f"cuda:{self.device}"
+        pbb = np.zeros(len(packs) + 1, dtype=np.uint64)
+        blob_base = 0
+        all_bases, all_nums, all_offs, all_blobs, totals = [], [], [], [], []
+        for i, (bases, nums, offs, blob, total) in enumerate(packs):
+            pbb[i + 1] = pbb[i] + bases.size
+            all_bases.append(bases)
+            all_nums.append(nums)
+            all_offs.append(np.asarray(offs[:-1], dtype=np.uint64) + np.uint64(blob_base))
+            all_blobs.append(blob)
+            totals.append(total)
+            blob_base += blob.size
+        cat = lambda xs, dt: np.ascontiguousarray(
+            np.concatenate(xs) if xs else np.empty(0, dtype=dt), dtype=dt)
+        bases = cat(all_bases, np.uint64)
+        nums = cat(all_nums, np.uint32)
+        offs = np.concatenate([cat(all_offs, np.uint64),
+                               np.array([blob_base], dtype=np.uint64)])
+        blob = cat(all_blobs, np.uint8)
+        dp = self.upload_pack(bases, nums, offs, blob, 0)
+        dp.pbb = pbb
+        dp.pack_totals = totals
+        return dp
+
+    def intersect_packed_batch(self, dpb, vs, outs=None, afters=None):
+        """Batched algo.IntersectCompressedWith: every pack of the fan-out in
+        ONE grid.  vs: per-pack CUDA int64 tensors (may repeat one shared
+        tensor — the q.UidList shape).  Returns (outs, lens)."""
+        import torch
+        from dgraph_amd._lib import UaPTask
+        n = len(dpb.pbb) - 1
+        assert len(vs) == n
+        if afters is None:
+            afters = [0] * n
+        if outs is None:
+            outs = [torch.empty(max(min(dpb.pack_totals[i], vs[i].numel()), 1),
+                                dtype=torch.int64, device=vs[i].device)
+                    for i in range(n)]
+        tasks = (UaPTask * n)()
+        for i in range(n):
+            tasks[i].v = vs[i].data_ptr()
+            tasks[i].m = vs[i].numel()
+            tasks[i].out = outs[i].data_ptr()
+            tasks[i].after_uid = afters[i]
+        pbb = (C.c_uint64 * (n + 1))(*[int(x) for x in dpb.pbb])
+        lens = (C.c_uint64 * n)()
+        check(lib().ua_intersect_packed_batch_dev(
+            self._ctx, C.c_void_p(dpb.bases.data_ptr()),
+            C.c_void_p(dpb.num_uids.data_ptr()), C.c_void_p(dpb.delta_offs.data_ptr()),
+            C.c_void_p(dpb.deltas.data_ptr()), pbb, n, tasks, lens))
+        return outs, [int(lens[i]) for i in range(n)]
+
     def intersect_packed(self, dpack, after, v, out=None):
         """algo.IntersectCompressedWith (uidlist.go:33): fused decode+intersect."""
         import torch

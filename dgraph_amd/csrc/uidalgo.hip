@@ -634,12 +634,25 @@ __device__ __forceinline__ u32 d_load_le(const u8 *p, int len) {
     return x;
 }
 
+/* per-block pack lookup for the batched form */
+__device__ __forceinline__ int d_pack_of(const u64 *__restrict__ pbb, int n_packs, u64 b) {
+    int lo = 0, hi = n_packs - 1;
+    while (lo < hi) {
+        int mid = (lo + hi + 1) >> 1;
+        if (pbb[mid] <= b) lo = mid;
+        else hi = mid - 1;
+    }
+    return lo;
+}
+
 template <int DECODE_ONLY>
 __global__ __launch_bounds__(UA_BLOCK) void k_packed(
     const u64 *__restrict__ bases, const u32 *__restrict__ nums,
     const u64 *__restrict__ doffs, const u8 *__restrict__ deltas, u64 n_blocks,
-    u64 after, const u64 *__restrict__ v, u64 m, u64 *__restrict__ staging,
-    u32 *__restrict__ blk_cnt) {
+    u64 after, const u64 *v, u64 m, u64 *__restrict__ staging,
+    u32 *__restrict__ blk_cnt,
+    const u64 *__restrict__ pbb /* batch: [n_packs+1] or null */, int n_packs,
+    const ua_ptask *__restrict__ tasks /* batch: per-pack v/m/out/after */) {
     __shared__ u8 sdel[UA_PKW][UA_MAX_DELTAS];
     __shared__ u64 sdec[UA_PKW][UA_MAX_BLOCK_UIDS + 4];
     __shared__ u16 sgoff[UA_PKW][64];
@@ -662,6 +675,12 @@ __global__ __launch_bounds__(UA_BLOCK) void k_packed(
             if (lane == 0) blk_cnt[b] = 0;
             active = false;
         }
+    }
+    if (active && tasks != nullptr) {
+        ua_ptask tk = tasks[d_pack_of(pbb, n_packs, b)];
+        after = tk.after_uid;
+        v = tk.v;
+        m = tk.m;
     }
     if (active) {
         for (u32 i = lane; i < dlen; i += 64) sdel[wv][i] = deltas[off0 + i];
@@ -755,6 +774,26 @@ __global__ __launch_bounds__(UA_BLOCK) void k_packed(
         }
     }
     if (lane == 0) blk_cnt[b] = cnt;
+}
+
+/* compact for the multi-pack batch: per-pack out destinations */
+__global__ __launch_bounds__(UA_BLOCK) void k_compact_pack(
+    const ua_ptask *__restrict__ tasks, const u64 *__restrict__ pbb, int n_packs,
+    const u32 *__restrict__ cnts, const u64 *__restrict__ offs,
+    const u64 *__restrict__ partials, const u64 *__restrict__ staging, u64 n_blocks) {
+    u64 basei = ((u64)blockIdx.x * 4 + (threadIdx.x >> 6)) * 4;
+    int lane = threadIdx.x & 63;
+    for (int q = 0; q < 4; q++) {
+        u64 b = basei + q;
+        if (b >= n_blocks) return;
+        u32 cnt = cnts[b];
+        if (cnt == 0) continue;
+        int p = d_pack_of(pbb, n_packs, b);
+        u64 *dst = tasks[p].out +
+                   (d_off(offs, partials, b) - d_off(offs, partials, pbb[p]));
+        const u64 *src = staging + b * (u64)UA_MAX_BLOCK_UIDS;
+        for (u32 i = lane; i < cnt; i += 64) dst[i] = src[i];
+    }
 }
 
 /* ==================== kernels: GPU codec.Encode ====================
@@ -1509,11 +1548,13 @@ static int run_packed_locked(ua_ctx *c, const ua_dpack *pk, u64 after, const u64
     if (decode_only) {
         hipLaunchKernelGGL(k_packed<1>, dim3((u32)nwg), dim3(UA_BLOCK), 0, c->stream,
                            pk->bases, pk->num_uids, pk->delta_offs, pk->deltas, nb, after,
-                           v, m, d_stage, d_cnt);
+                           v, m, d_stage, d_cnt, (const u64 *)nullptr, 0,
+                           (const ua_ptask *)nullptr);
     } else {
         hipLaunchKernelGGL(k_packed<0>, dim3((u32)nwg), dim3(UA_BLOCK), 0, c->stream,
                            pk->bases, pk->num_uids, pk->delta_offs, pk->deltas, nb, after,
-                           v, m, d_stage, d_cnt);
+                           v, m, d_stage, d_cnt, (const u64 *)nullptr, 0,
+                           (const ua_ptask *)nullptr);
     }
     HIP_TRY(hipEventRecord(c->ev[1], c->stream));
 
@@ -1550,6 +1591,84 @@ extern "C" int ua_intersect_packed_dev(ua_ctx *c, const ua_dpack *pk, uint64_t a
                                        uint64_t *out_n) {
     std::lock_guard<std::mutex> g(c->mu);
     return run_packed_locked(c, pk, after_uid, v, m, out, out_n, 0);
+}
+
+/* ---- multi-pack fan-out: one grid over every pack's blocks ---- */
+extern "C" int ua_intersect_packed_batch_dev(ua_ctx *c, const uint64_t *bases,
+                                             const uint32_t *num_uids,
+                                             const uint64_t *delta_offs,
+                                             const uint8_t *deltas,
+                                             const uint64_t *pack_block_base,
+                                             int n_packs, const ua_ptask *tasks,
+                                             uint64_t *out_lens) {
+    std::lock_guard<std::mutex> g(c->mu);
+    HIP_TRY(hipSetDevice(c->device));
+    if (n_packs <= 0) return UA_OK;
+    u64 nb = pack_block_base[n_packs];
+    int rc;
+    /* upload pbb + tasks in one copy (WS_DESC is free during this call) */
+    size_t pbb_bytes = ((size_t)n_packs + 1) * sizeof(u64);
+    size_t task_bytes = (size_t)n_packs * sizeof(ua_ptask);
+    if ((rc = ws_reserve(c, WS_DESC, pbb_bytes + task_bytes))) return rc;
+    u64 *d_pbb = (u64 *)c->ws[WS_DESC];
+    ua_ptask *d_tasks = (ua_ptask *)((u8 *)c->ws[WS_DESC] + pbb_bytes);
+    std::vector<u8> hostbuf(pbb_bytes + task_bytes);
+    memcpy(hostbuf.data(), pack_block_base, pbb_bytes);
+    memcpy(hostbuf.data() + pbb_bytes, tasks, task_bytes);
+    HIP_TRY(hipMemcpyAsync(c->ws[WS_DESC], hostbuf.data(), hostbuf.size(),
+                           hipMemcpyHostToDevice, c->stream));
+
+    if ((rc = ws_reserve(c, WS_TCNT, (nb + 1) * sizeof(u32)))) return rc;
+    if ((rc = ws_reserve(c, WS_TOFF, (nb + 1) * sizeof(u64)))) return rc;
+    if ((rc = ws_reserve(c, WS_STAGE, (nb ? nb : 1) * UA_MAX_BLOCK_UIDS * sizeof(u64))))
+        return rc;
+    if ((rc = ws_reserve(c, WS_POUT, (size_t)n_packs * sizeof(u64)))) return rc;
+    u32 *d_cnt = (u32 *)c->ws[WS_TCNT];
+    u64 *d_off_arr = (u64 *)c->ws[WS_TOFF];
+    u64 *d_stage = (u64 *)c->ws[WS_STAGE];
+    u64 *d_pout = (u64 *)c->ws[WS_POUT];
+    HIP_TRY(hipMemsetAsync(d_cnt + nb, 0, sizeof(u32), c->stream));
+
+    if (nb > 0) {
+        u64 nwg = (nb + UA_PKW - 1) / UA_PKW;
+        HIP_TRY(hipEventRecord(c->ev[0], c->stream));
+        hipLaunchKernelGGL(k_packed<0>, dim3((u32)nwg), dim3(UA_BLOCK), 0, c->stream,
+                           bases, num_uids, delta_offs, deltas, nb, (u64)0,
+                           (const u64 *)nullptr, (u64)0, d_stage, d_cnt, d_pbb,
+                           n_packs, d_tasks);
+        HIP_TRY(hipEventRecord(c->ev[1], c->stream));
+    }
+    if ((rc = run_scan(c, d_cnt, nb + 1, d_off_arr))) return rc;
+    u64 *d_part = (u64 *)c->ws[WS_PARTIAL];
+    if (nb > 0) {
+        hipLaunchKernelGGL(k_compact_pack, dim3((u32)((nb + 15) / 16)), dim3(UA_BLOCK),
+                           0, c->stream, d_tasks, d_pbb, n_packs, d_cnt, d_off_arr,
+                           d_part, d_stage, nb);
+    }
+    u64 poutblk = ((u64)n_packs + UA_BLOCK - 1) / UA_BLOCK;
+    hipLaunchKernelGGL(k_pair_out, dim3((u32)poutblk), dim3(UA_BLOCK), 0, c->stream,
+                       d_off_arr, d_part, d_pbb, n_packs, d_pout);
+    HIP_TRY(hipMemcpyAsync(out_lens, d_pout, (size_t)n_packs * sizeof(u64),
+                           hipMemcpyDeviceToHost, c->stream));
+    u64 dbytes = 0;
+    HIP_TRY(hipMemcpyAsync(&dbytes, delta_offs + nb, sizeof(u64),
+                           hipMemcpyDeviceToHost, c->stream));
+    HIP_TRY(hipStreamSynchronize(c->stream));
+    HIP_TRY(hipGetLastError());
+
+    if (nb > 0) {
+        float ms = 0.f;
+        HIP_TRY(hipEventElapsedTime(&ms, c->ev[0], c->ev[1]));
+        c->kernel_ms += ms;
+        c->n_launches += 1;
+    }
+    u64 out_elems = 0, m_elems = 0;
+    for (int p = 0; p < n_packs; p++) {
+        out_elems += out_lens[p];
+        m_elems += tasks[p].m;
+    }
+    c->bytes_algo += dbytes + nb * 20 + 8 * (m_elems + out_elems);
+    return UA_OK;
 }
 
 extern "C" int ua_decode_dev(ua_ctx *c, const ua_dpack *pk, uint64_t seek_uid, uint64_t *out,

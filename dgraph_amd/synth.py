@@ -6,9 +6,12 @@ SEED = 0xD6A77
 
 
 def gen_sorted_unique(rng, n, limit):
-    """n distinct sorted values uniform over [0, limit)."""
+    """n distinct sorted values uniform over [0, limit).  limit is widened to
+    2n when the range is too small to hold n distinct values."""
     if n == 0:
         return np.empty(0, dtype=np.uint64)
+    if limit < 2 * n:
+        limit = 2 * n
     draw = rng.integers(0, limit, size=int(n * 1.3) + 16, dtype=np.uint64)
     un = np.unique(draw)
     while un.size < n:

@@ -173,6 +173,27 @@ int ua_intersect_packed_dev(ua_ctx *, const ua_dpack *, uint64_t after_uid,
                             const uint64_t *v, uint64_t m, uint64_t *out,
                             uint64_t *out_n);
 
+/* Batched fused decode+intersect — ONE grid over a whole per-key fan-out
+ * (worker/task.go:834-971 handleUidPostings: thousands of keys' packs, each
+ * intersected with its filter list, often one shared q.UidList).  All packs
+ * are concatenated into one flat block arena (bases/num_uids/delta_offs/
+ * deltas, like ua_dpack); pack_block_base[n_packs+1] delimits each pack's
+ * blocks.  Per-pack v/out/after are device pointers in the tasks array. */
+typedef struct {
+    const uint64_t *v; /* device */
+    uint64_t m;
+    uint64_t *out;     /* device, capacity >= min(pack uids, m) */
+    uint64_t after_uid;
+} ua_ptask;
+
+int ua_intersect_packed_batch_dev(ua_ctx *, const uint64_t *bases,
+                                  const uint32_t *num_uids,
+                                  const uint64_t *delta_offs,
+                                  const uint8_t *deltas,
+                                  const uint64_t *pack_block_base /* host [n_packs+1] */,
+                                  int n_packs, const ua_ptask *tasks /* host */,
+                                  uint64_t *out_lens);
+
 /* ---- host-pointer convenience (upload+compute+download; mirrors the algo
  * package signatures 1:1 for the cgo shim) ---- */
 int ua_intersect(ua_ctx *, const uint64_t *u, uint64_t n, const uint64_t *v,

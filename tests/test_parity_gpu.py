@@ -361,6 +361,44 @@ def test_packed_compositions_vs_oracle(eng):
     assert to_np(eng.decode_pack(fp)).tolist() == want_f.tolist()
 
 
+def test_packed_fanout_batch_vs_oracle(eng):
+    """The handleUidPostings fan-out shape (worker/task.go:834-971): many
+    keys' packs intersected in ONE grid — half against one shared q.UidList,
+    half against per-key lists, with per-pack after cursors."""
+    rng = np.random.default_rng(SEED + 99)
+    n_packs = 64
+    packs_np, flat, vs_np, afters = [], [], [], []
+    shared = synth.gen_sorted_unique(rng, 50_000, 2_000_000)
+    d_shared = to_dev(shared)
+    d_vs = []
+    bss = []
+    for i in range(n_packs):
+        size = int(rng.integers(1, 60_000))
+        uids = np.unique(synth.getuids_geometric(rng, size))
+        packs_np.append(uids)
+        bs = int(rng.choice([10, 64, 256]))
+        bss.append(bs)
+        flat.append(algo.encode_flat(uids, bs))
+        if i % 2 == 0:
+            vs_np.append(shared)
+            d_vs.append(d_shared)
+        else:
+            mlen = int(rng.integers(1, 30_000))
+            v = synth.gen_sorted_unique(
+                rng, mlen, max(int(uids[-1]) + 1000 if uids.size else 1000, 3 * mlen))
+            vs_np.append(v)
+            d_vs.append(to_dev(v))
+        afters.append(0 if i % 4 else int(uids[uids.size // 2]) if uids.size else 0)
+
+    dpb = eng.upload_pack_batch(flat)
+    outs, lens = eng.intersect_packed_batch(dpb, d_vs, afters=afters)
+    for i in range(n_packs):
+        got = to_np(outs[i][:lens[i]])
+        opack = orc.Pack(packs_np[i], bss[i])
+        want = orc.intersect_compressed_with(opack, afters[i], vs_np[i])
+        assert got.tolist() == want.tolist(), f"pack {i}"
+
+
 def test_host_packed_api(eng):
     """ua_intersect_packed host-pointer convenience (the cgo surface)."""
     import ctypes as Ct
