@@ -506,6 +506,166 @@ __global__ __launch_bounds__(UA_BLOCK) void k_tiles(
     }
 }
 
+/* ---- 2-tile software-pipelined variant: one workgroup processes tiles
+ * (2w, 2w+1); the global loads for tile 2w+1 are issued into registers
+ * before the walk of tile 2w, so HBM latency hides under compute.  Double
+ * LDS buffer (2x16.4KB -> 4 WGs/CU).  Toggle: UA_PIPE. ---- */
+
+struct TileMeta {
+    u32 p;
+    u32 a0, b0;
+    int alen, blen;
+    bool valid, has_ab, has_bn;
+    UaDesc d;
+};
+
+__device__ __forceinline__ TileMeta d_tile_meta(const UaDesc *__restrict__ descs,
+                                                const u32 *__restrict__ tile_pair,
+                                                const u32 *__restrict__ tile_a0,
+                                                u64 total_tiles, u64 t) {
+    TileMeta m;
+    m.valid = (t < total_tiles);
+    if (!m.valid) {
+        m.alen = m.blen = 0;
+        m.a0 = m.b0 = 0;
+        m.has_ab = m.has_bn = false;
+        return m;
+    }
+    m.p = tile_pair[t];
+    m.d = descs[m.p];
+    u64 lt = t - m.d.tile_base;
+    u64 path = m.d.n + m.d.m;
+    u64 d0 = lt * UA_TILE;
+    u64 d1 = d0 + UA_TILE;
+    if (d1 > path) d1 = path;
+    u32 a1 = (t + 1 < total_tiles && tile_pair[t + 1] == m.p) ? tile_a0[t + 1]
+                                                              : (u32)m.d.n;
+    m.a0 = tile_a0[t];
+    m.b0 = (u32)(d0 - m.a0);
+    u32 b1 = (u32)(d1 - a1);
+    m.alen = (int)(a1 - m.a0);
+    m.blen = (int)(b1 - m.b0);
+    m.has_ab = (m.a0 > 0);
+    m.has_bn = ((u64)b1 < m.d.m);
+    return m;
+}
+
+/* unified A|B element load into registers (one code path, coalesced) */
+#define UA_PIPE_REGS (UA_WPT)
+__device__ __forceinline__ void d_reg_load(const TileMeta &m, int tid,
+                                           u64 (&val)[UA_PIPE_REGS], u64 &abefore,
+                                           u64 &bnext) {
+#pragma unroll
+    for (int k = 0; k < UA_PIPE_REGS; k++) {
+        int e = tid + k * UA_BLOCK;
+        u64 x = 0;
+        if (e < m.alen) x = m.d.u[m.a0 + e];
+        else if (e - m.alen < m.blen) x = m.d.v[m.b0 + (e - m.alen)];
+        val[k] = x;
+    }
+    if (tid == 0) {
+        abefore = m.has_ab ? m.d.u[m.a0 - 1] : 0;
+        bnext = m.has_bn ? m.d.v[m.b0 + m.blen] : 0;
+    }
+}
+
+__device__ __forceinline__ void d_reg_commit(const TileMeta &m, int tid, u64 *smem,
+                                             const u64 (&val)[UA_PIPE_REGS],
+                                             u64 abefore, u64 bnext, u64 *s_abefore) {
+    int boff = (m.alen + 1) & ~1;
+#pragma unroll
+    for (int k = 0; k < UA_PIPE_REGS; k++) {
+        int e = tid + k * UA_BLOCK;
+        if (e < m.alen) smem[e] = val[k];
+        else if (e - m.alen < m.blen) smem[boff + (e - m.alen)] = val[k];
+    }
+    if (tid == 0) {
+        *s_abefore = abefore;
+        smem[boff + m.blen] = bnext; /* lookahead slot (garbage if !has_bn) */
+    }
+}
+
+template <int OP, int MODE>
+__device__ __forceinline__ void d_tile_body(const TileMeta &m, u64 t, int tid,
+                                            const u64 *smem, u64 a_before,
+                                            u64 *__restrict__ staging, u64 stage_stride,
+                                            u32 *__restrict__ tile_cnt,
+                                            const u64 *__restrict__ offs,
+                                            const u64 *__restrict__ partials,
+                                            u32 *scanbuf) {
+    int alen = m.alen, blen = m.blen;
+    int boff = (alen + 1) & ~1;
+    int tilelen = alen + blen;
+    int s0 = tid * UA_WPT;
+    int s1 = s0 + UA_WPT;
+    if (s0 > tilelen) s0 = tilelen;
+    if (s1 > tilelen) s1 = tilelen;
+    u64 em[UA_WPT];
+    u32 flags = 0;
+    int cnt = 0;
+    if (m.valid) {
+        int i0 = d_merge_path_lds(smem, alen, smem + boff, blen, s0);
+        cnt = tile_walk2<OP>(smem, 0, alen, boff, blen, a_before, m.has_ab, m.has_bn,
+                             s0, s1, i0, em, flags);
+    }
+    u32 excl, total;
+    d_block_scan(tid, (u32)cnt, scanbuf, excl, total);
+    if (!m.valid) return;
+    if (MODE == MODE_COUNT) {
+        if (tid == 0) tile_cnt[t] = total;
+        return;
+    }
+    u64 *dst;
+    if (MODE == MODE_STAGE) {
+        dst = staging + t * stage_stride + excl;
+        if (tid == 0) tile_cnt[t] = total;
+    } else {
+        dst = m.d.out + (d_off(offs, partials, t) - d_off(offs, partials, m.d.tile_base)) +
+              excl;
+    }
+    if (cnt > 0) {
+        int k = 0;
+#pragma unroll
+        for (int s = 0; s < UA_WPT; s++) {
+            if (flags & (1u << s)) dst[k++] = em[s];
+        }
+    }
+}
+
+template <int OP, int MODE>
+__global__ __launch_bounds__(UA_BLOCK) void k_tiles_pipe(
+    const UaDesc *__restrict__ descs, const u32 *__restrict__ tile_pair,
+    const u32 *__restrict__ tile_a0, u64 total_tiles,
+    u64 *__restrict__ staging, u64 stage_stride, u32 *__restrict__ tile_cnt,
+    const u64 *__restrict__ offs, const u64 *__restrict__ partials) {
+    __shared__ __align__(16) u64 smem[2][UA_TILE + 4];
+    __shared__ u32 scanbuf[2][UA_BLOCK / 64];
+    __shared__ u64 s_abefore[2];
+
+    int tid = threadIdx.x;
+    u64 t0 = (u64)blockIdx.x * 2;
+    u64 t1 = t0 + 1;
+    TileMeta m0 = d_tile_meta(descs, tile_pair, tile_a0, total_tiles, t0);
+    TileMeta m1 = d_tile_meta(descs, tile_pair, tile_a0, total_tiles, t1);
+
+    u64 v0[UA_PIPE_REGS], ab0 = 0, bn0 = 0;
+    d_reg_load(m0, tid, v0, ab0, bn0);
+    d_reg_commit(m0, tid, smem[0], v0, ab0, bn0, &s_abefore[0]);
+    /* issue tile-1 loads NOW: they stay in flight across the barrier and
+     * the tile-0 walk */
+    u64 v1[UA_PIPE_REGS], ab1 = 0, bn1 = 0;
+    if (m1.valid) d_reg_load(m1, tid, v1, ab1, bn1);
+    __syncthreads();
+
+    d_tile_body<OP, MODE>(m0, t0, tid, smem[0], s_abefore[0], staging, stage_stride,
+                          tile_cnt, offs, partials, scanbuf[0]);
+    if (!m1.valid) return;
+    d_reg_commit(m1, tid, smem[1], v1, ab1, bn1, &s_abefore[1]);
+    __syncthreads();
+    d_tile_body<OP, MODE>(m1, t1, tid, smem[1], s_abefore[1], staging, stage_stride,
+                          tile_cnt, offs, partials, scanbuf[1]);
+}
+
 /* ==================== kernels: flat hierarchical scan (u32 -> u64) ==================== */
 
 __global__ __launch_bounds__(UA_BLOCK) void k_scan1(const u32 *__restrict__ cnt, u64 n,
@@ -579,10 +739,15 @@ __global__ __launch_bounds__(UA_BLOCK) void k_compact(
     u64 stage_stride, u64 total_tiles) {
     u64 base = ((u64)blockIdx.x * 4 + (threadIdx.x >> 6)) * 4;
     int lane = threadIdx.x & 63;
+    if (base >= total_tiles) return;
+    u32 cnts4[4]; /* prefetch: 4 independent loads instead of a dependent chain */
+#pragma unroll
+    for (int q = 0; q < 4; q++)
+        cnts4[q] = (base + q < total_tiles) ? tile_cnt[base + q] : 0;
+#pragma unroll
     for (int q = 0; q < 4; q++) {
         u64 t = base + q;
-        if (t >= total_tiles) return;
-        u32 cnt = tile_cnt[t];
+        u32 cnt = cnts4[q];
         if (cnt == 0) continue;
         u32 p = tile_pair[t];
         UaDesc d = descs[p];
@@ -598,10 +763,14 @@ __global__ __launch_bounds__(UA_BLOCK) void k_compact_flat(
     u64 stage_stride, u64 n_blocks) {
     u64 base = ((u64)blockIdx.x * 4 + (threadIdx.x >> 6)) * 4;
     int lane = threadIdx.x & 63;
+    if (base >= n_blocks) return;
+    u32 cnts4[4];
+#pragma unroll
+    for (int q = 0; q < 4; q++) cnts4[q] = (base + q < n_blocks) ? cnts[base + q] : 0;
+#pragma unroll
     for (int q = 0; q < 4; q++) {
         u64 b = base + q;
-        if (b >= n_blocks) return;
-        u32 cnt = cnts[b];
+        u32 cnt = cnts4[q];
         if (cnt == 0) continue;
         u64 *dst = out + d_off(offs, partials, b);
         const u64 *src = staging + b * stage_stride;
@@ -783,10 +952,14 @@ __global__ __launch_bounds__(UA_BLOCK) void k_compact_pack(
     const u64 *__restrict__ partials, const u64 *__restrict__ staging, u64 n_blocks) {
     u64 basei = ((u64)blockIdx.x * 4 + (threadIdx.x >> 6)) * 4;
     int lane = threadIdx.x & 63;
+    if (basei >= n_blocks) return;
+    u32 cnts4[4];
+#pragma unroll
+    for (int q = 0; q < 4; q++) cnts4[q] = (basei + q < n_blocks) ? cnts[basei + q] : 0;
+#pragma unroll
     for (int q = 0; q < 4; q++) {
         u64 b = basei + q;
-        if (b >= n_blocks) return;
-        u32 cnt = cnts[b];
+        u32 cnt = cnts4[q];
         if (cnt == 0) continue;
         int p = d_pack_of(pbb, n_packs, b);
         u64 *dst = tasks[p].out +
@@ -923,6 +1096,16 @@ __global__ __launch_bounds__(UA_BLOCK) void k_enc_finalize(
     u8 *dst = deltas + off;
     for (u32 i = lane; i < bytes; i += 64) dst[i] = src[i];
 }
+
+
+#ifndef UA_PIPE
+#define UA_PIPE 0 /* 1 = 2-tile software-pipelined tile kernel (A/B toggle) */
+#endif
+
+template <int OP, int MODE>
+static void launch_tiles(ua_ctx *c, const UaDesc *descs, const u32 *tpair,
+                         const u32 *ta0, u64 T, u64 *stage, u64 stride, u32 *tcnt,
+                         const u64 *offs, const u64 *part);
 
 /* ==================== host shim ==================== */
 
@@ -1066,6 +1249,20 @@ extern "C" int ua_stats_get(ua_ctx *c, uint64_t *n_launches, double *kernel_ms,
 
 /* ---- flat scan helper: cnt u32[n] (+1 zero sentinel at n-1 position
  * provided by caller) -> offs u64[n] exclusive scan ---- */
+template <int OP, int MODE>
+static void launch_tiles(ua_ctx *c, const UaDesc *descs, const u32 *tpair,
+                         const u32 *ta0, u64 T, u64 *stage, u64 stride, u32 *tcnt,
+                         const u64 *offs, const u64 *part) {
+#if UA_PIPE
+    hipLaunchKernelGGL((k_tiles_pipe<OP, MODE>), dim3((u32)((T + 1) / 2)),
+                       dim3(UA_BLOCK), 0, c->stream, descs, tpair, ta0, T, stage,
+                       stride, tcnt, offs, part);
+#else
+    hipLaunchKernelGGL((k_tiles<OP, MODE>), dim3((u32)T), dim3(UA_BLOCK), 0, c->stream,
+                       descs, tpair, ta0, T, stage, stride, tcnt, offs, part);
+#endif
+}
+
 /* Split flat scan: offs_dev gets chunk-local exclusive offsets, WS_PARTIAL
  * gets the scanned chunk partials; consumers combine via d_off(). */
 static int run_scan(ua_ctx *c, const u32 *cnt_dev, u64 n, u64 *offs_dev) {
@@ -1142,20 +1339,16 @@ static int run_batch_locked(ua_ctx *c, const ua_dpair *pairs, int n_pairs,
 
         HIP_TRY(hipEventRecord(c->ev[0], c->stream));
         if (op == OP_INTERSECT) {
-            hipLaunchKernelGGL((k_tiles<OP_INTERSECT, MODE_STAGE>), dim3((u32)total_tiles),
-                               dim3(UA_BLOCK), 0, c->stream, d_descs, d_tpair, d_ta0,
-                               total_tiles, d_stage, stage_stride, d_tcnt, (u64 *)nullptr,
-                               (u64 *)nullptr);
+            launch_tiles<OP_INTERSECT, MODE_STAGE>(c, d_descs, d_tpair, d_ta0,
+                                                   total_tiles, d_stage, stage_stride,
+                                                   d_tcnt, nullptr, nullptr);
         } else if (op == OP_DIFF) {
-            hipLaunchKernelGGL((k_tiles<OP_DIFF, MODE_STAGE>), dim3((u32)total_tiles),
-                               dim3(UA_BLOCK), 0, c->stream, d_descs, d_tpair, d_ta0,
-                               total_tiles, d_stage, stage_stride, d_tcnt, (u64 *)nullptr,
-                               (u64 *)nullptr);
+            launch_tiles<OP_DIFF, MODE_STAGE>(c, d_descs, d_tpair, d_ta0, total_tiles,
+                                              d_stage, stage_stride, d_tcnt, nullptr,
+                                              nullptr);
         } else {
-            hipLaunchKernelGGL((k_tiles<OP_UNION, MODE_COUNT>), dim3((u32)total_tiles),
-                               dim3(UA_BLOCK), 0, c->stream, d_descs, d_tpair, d_ta0,
-                               total_tiles, (u64 *)nullptr, 0, d_tcnt, (u64 *)nullptr,
-                               (u64 *)nullptr);
+            launch_tiles<OP_UNION, MODE_COUNT>(c, d_descs, d_tpair, d_ta0, total_tiles,
+                                               nullptr, 0, d_tcnt, nullptr, nullptr);
         }
         HIP_TRY(hipEventRecord(c->ev[1], c->stream));
 
@@ -1164,9 +1357,8 @@ static int run_batch_locked(ua_ctx *c, const ua_dpair *pairs, int n_pairs,
 
         if (op == OP_UNION) {
             HIP_TRY(hipEventRecord(c->ev[2], c->stream));
-            hipLaunchKernelGGL((k_tiles<OP_UNION, MODE_WRITE>), dim3((u32)total_tiles),
-                               dim3(UA_BLOCK), 0, c->stream, d_descs, d_tpair, d_ta0,
-                               total_tiles, (u64 *)nullptr, 0, d_tcnt, d_toff, d_part);
+            launch_tiles<OP_UNION, MODE_WRITE>(c, d_descs, d_tpair, d_ta0, total_tiles,
+                                               nullptr, 0, d_tcnt, d_toff, d_part);
             HIP_TRY(hipEventRecord(c->ev[3], c->stream));
         } else {
             u64 cblk = (total_tiles + 15) / 16;
@@ -1338,20 +1530,16 @@ extern "C" int ua_batch_run(ua_ctx *c, ua_batch *b, int op, uint64_t *out_lens) 
     if (T > 0) {
         HIP_TRY(hipEventRecord(c->ev[0], c->stream));
         if (kop == OP_INTERSECT) {
-            hipLaunchKernelGGL((k_tiles<OP_INTERSECT, MODE_STAGE>), dim3((u32)T),
-                               dim3(UA_BLOCK), 0, c->stream, b->d_descs, b->d_tpair,
-                               b->d_ta0, T, b->d_stage, stride, b->d_tcnt,
-                               (u64 *)nullptr, (u64 *)nullptr);
+            launch_tiles<OP_INTERSECT, MODE_STAGE>(c, b->d_descs, b->d_tpair, b->d_ta0,
+                                                   T, b->d_stage, stride, b->d_tcnt,
+                                                   nullptr, nullptr);
         } else if (kop == OP_DIFF) {
-            hipLaunchKernelGGL((k_tiles<OP_DIFF, MODE_STAGE>), dim3((u32)T),
-                               dim3(UA_BLOCK), 0, c->stream, b->d_descs, b->d_tpair,
-                               b->d_ta0, T, b->d_stage, stride, b->d_tcnt,
-                               (u64 *)nullptr, (u64 *)nullptr);
+            launch_tiles<OP_DIFF, MODE_STAGE>(c, b->d_descs, b->d_tpair, b->d_ta0, T,
+                                              b->d_stage, stride, b->d_tcnt, nullptr,
+                                              nullptr);
         } else {
-            hipLaunchKernelGGL((k_tiles<OP_UNION, MODE_COUNT>), dim3((u32)T),
-                               dim3(UA_BLOCK), 0, c->stream, b->d_descs, b->d_tpair,
-                               b->d_ta0, T, (u64 *)nullptr, 0, b->d_tcnt,
-                               (u64 *)nullptr, (u64 *)nullptr);
+            launch_tiles<OP_UNION, MODE_COUNT>(c, b->d_descs, b->d_tpair, b->d_ta0, T,
+                                               nullptr, 0, b->d_tcnt, nullptr, nullptr);
         }
         HIP_TRY(hipEventRecord(c->ev[1], c->stream));
 
@@ -1361,10 +1549,9 @@ extern "C" int ua_batch_run(ua_ctx *c, ua_batch *b, int op, uint64_t *out_lens) 
                            b->nchunks);
         if (kop == OP_UNION) {
             HIP_TRY(hipEventRecord(c->ev[2], c->stream));
-            hipLaunchKernelGGL((k_tiles<OP_UNION, MODE_WRITE>), dim3((u32)T),
-                               dim3(UA_BLOCK), 0, c->stream, b->d_descs, b->d_tpair,
-                               b->d_ta0, T, (u64 *)nullptr, 0, b->d_tcnt, b->d_toff,
-                               b->d_part);
+            launch_tiles<OP_UNION, MODE_WRITE>(c, b->d_descs, b->d_tpair, b->d_ta0, T,
+                                               nullptr, 0, b->d_tcnt, b->d_toff,
+                                               b->d_part);
             HIP_TRY(hipEventRecord(c->ev[3], c->stream));
         } else {
             hipLaunchKernelGGL(k_compact, dim3((u32)((T + 15) / 16)), dim3(UA_BLOCK), 0,
