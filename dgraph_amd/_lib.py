@@ -42,6 +42,14 @@ class UaDPack(C.Structure):
     ]
 
 
+class UaDSeg(C.Structure):
+    _fields_ = [
+        ("data", C.c_void_p),
+        ("n", _u64),
+        ("tmp", C.c_void_p),
+    ]
+
+
 class UaPTask(C.Structure):
     _fields_ = [
         ("v", C.c_void_p),
@@ -99,6 +107,8 @@ def lib():
         L.ua_batch_destroy.restype = None
         L.ua_merge_batch_dev.argtypes = [C.c_void_p, C.POINTER(UaDPair), C.c_int, _u64p]
         L.ua_difference_batch_dev.argtypes = [C.c_void_p, C.POINTER(UaDPair), C.c_int, _u64p]
+        L.ua_merge_all_batch_dev.argtypes = [C.c_void_p, C.POINTER(UaDPair), C.c_int, _u64p]
+        L.ua_sort_segments_dev.argtypes = [C.c_void_p, C.POINTER(UaDSeg), C.c_int]
         L.ua_index_of_batch_dev.argtypes = [C.c_void_p, C.c_void_p, _u64, C.c_void_p, _u64,
                                             C.c_void_p]
         L.ua_intersect_k_dev.argtypes = [C.c_void_p, _voidpp, _u64p, C.c_int, C.c_void_p, _u64p]

@@ -156,6 +156,34 @@ class Engine:
                                    lens, k, C.c_void_p(out.data_ptr()), C.byref(out_n)))
         return out[:out_n.value]
 
+    def merge_all_pairs(self, us, vs, outs=None):
+        """Batched duplicate-KEEPING merge of sorted runs (no dedup)."""
+        import torch
+        if outs is None:
+            outs = [torch.empty(max(u.numel() + v.numel(), 1), dtype=torch.int64,
+                                device=u.device) for u, v in zip(us, vs)]
+        lens = self._run_pairs(lib().ua_merge_all_batch_dev, us, vs, outs)
+        return outs, lens
+
+    def sort_segments(self, tensors):
+        """Batched segmented sort in place (u64 ascending, duplicates kept):
+        the sort-path primitive (worker/sort.go UidMatrix shapes, SURVEY
+        §8f row 3)."""
+        import torch
+        from dgraph_amd._lib import UaDSeg
+        k = len(tensors)
+        if k == 0:
+            return tensors
+        tmps = [torch.empty(max(t.numel(), 1), dtype=torch.int64, device=t.device)
+                for t in tensors]
+        segs = (UaDSeg * k)()
+        for i, t in enumerate(tensors):
+            segs[i].data = t.data_ptr()
+            segs[i].n = t.numel()
+            segs[i].tmp = tmps[i].data_ptr()
+        check(lib().ua_sort_segments_dev(self._ctx, segs, k))
+        return tensors
+
     def index_of_batch(self, u, queries):
         """Batched algo.IndexOf: u, queries CUDA int64; returns int64 tensor
         of positions (-1 = absent)."""

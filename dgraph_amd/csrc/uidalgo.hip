@@ -174,8 +174,8 @@ __global__ __launch_bounds__(UA_BLOCK) void k_partition(
 
 /* ==================== kernel: tile set-algebra ==================== */
 
-enum { OP_INTERSECT = 0, OP_UNION = 1, OP_DIFF = 2 };
-enum { MODE_STAGE = 0, MODE_COUNT = 1, MODE_WRITE = 2 };
+enum { OP_INTERSECT = 0, OP_UNION = 1, OP_DIFF = 2, OP_MERGE_ALL = 3 };
+enum { MODE_STAGE = 0, MODE_COUNT = 1, MODE_WRITE = 2, MODE_DIRECT = 3 };
 
 /* One pass over the thread's merge-path segment.  Emissions are recorded in
  * a statically-indexed register array (em[s], flag bit s) so the walk runs
@@ -354,6 +354,9 @@ __device__ __forceinline__ int tile_walk2(const u64 *smembase, int aoff, int ale
         } else if (OP == OP_DIFF) {
             emit = takeA && !eq;
             em[s] = a;
+        } else if (OP == OP_MERGE_ALL) { /* duplicate-keeping merge (sort runs) */
+            emit = true;
+            em[s] = takeA ? a : b;
         } else { /* UNION */
             bool dupB = (i > 0 || has_ab) && (b == prev_a);
             emit = takeA || !dupB;
@@ -457,7 +460,7 @@ __global__ __launch_bounds__(UA_BLOCK) void k_tiles(
 #ifndef UA_WALK2
 #define UA_WALK2 1 /* 0 = the branchy register-frontier walk */
 #endif
-    if (UA_SEARCH && OP != OP_UNION) {
+    if (UA_SEARCH && OP != OP_UNION && OP != OP_MERGE_ALL) {
         cnt = tile_search<OP>(As, alen, Bs, blen, has_bn, tid, em, flags);
     } else {
         int tilelen = alen + blen;
@@ -473,6 +476,17 @@ __global__ __launch_bounds__(UA_BLOCK) void k_tiles(
         } else {
             cnt = tile_walk<OP>(As, alen, Bs, blen, a_before, has_ab, has_bn, s0, s1,
                                 i0, em, flags);
+        }
+        if (MODE == MODE_DIRECT) {
+            /* OP_MERGE_ALL: every path element emits, so the output position
+             * IS the path position — no scan, no staging, no compaction */
+            u64 *dst = d.out + d0 + (u64)s0;
+            int steps = s1 - s0;
+#pragma unroll
+            for (int s = 0; s < UA_WPT; s++) {
+                if (s < steps) dst[s] = em[s];
+            }
+            return;
         }
     }
 #endif
@@ -664,6 +678,49 @@ __global__ __launch_bounds__(UA_BLOCK) void k_tiles_pipe(
     __syncthreads();
     d_tile_body<OP, MODE>(m1, t1, tid, smem[1], s_abefore[1], staging, stage_stride,
                           tile_cnt, offs, partials, scanbuf[1]);
+}
+
+/* ==================== kernel: bitonic chunk sort (segmented sort stage 1) ====================
+ * One workgroup sorts one <=2048-element chunk in LDS (u64 ascending,
+ * duplicates kept; padded with UINT64_MAX).  Stage 2 is the merge-path
+ * OP_MERGE_ALL tree. */
+
+#define UA_SORT_N 2048 /* bitonic chunk length: power of two, independent of UA_TILE */
+
+struct UaChunk {
+    const u64 *src;
+    u64 *dst;
+    u32 len;
+    u32 pad_;
+};
+
+__global__ __launch_bounds__(UA_BLOCK) void k_sort_chunks(
+    const UaChunk *__restrict__ chunks, u64 n_chunks) {
+    __shared__ u64 s[UA_SORT_N];
+    u64 cidx = blockIdx.x;
+    if (cidx >= n_chunks) return;
+    UaChunk ch = chunks[cidx];
+    int tid = threadIdx.x;
+    for (int i = tid; i < UA_SORT_N; i += UA_BLOCK)
+        s[i] = (i < (int)ch.len) ? ch.src[i] : UINT64_MAX;
+    __syncthreads();
+    for (int k = 2; k <= UA_SORT_N; k <<= 1) {
+        for (int j = k >> 1; j > 0; j >>= 1) {
+            for (int i = tid; i < UA_SORT_N; i += UA_BLOCK) {
+                int p = i ^ j;
+                if (p > i) {
+                    bool up = ((i & k) == 0);
+                    u64 x = s[i], y = s[p];
+                    if ((x > y) == up) {
+                        s[i] = y;
+                        s[p] = x;
+                    }
+                }
+            }
+            __syncthreads();
+        }
+    }
+    for (int i = tid; i < (int)ch.len; i += UA_BLOCK) ch.dst[i] = s[i];
 }
 
 /* ==================== kernels: flat hierarchical scan (u32 -> u64) ==================== */
@@ -1136,7 +1193,7 @@ struct ua_ctx {
     uint64_t n_launches = 0;
     double kernel_ms = 0.0;
     uint64_t bytes_algo = 0;
-    std::mutex mu;
+    std::recursive_mutex mu; /* compound ops hold it across internal stages */
 };
 
 extern "C" const char *ua_strerror(int code) {
@@ -1231,7 +1288,7 @@ extern "C" int ua_sync(ua_ctx *c) {
 }
 
 extern "C" int ua_stats_reset(ua_ctx *c) {
-    std::lock_guard<std::mutex> g(c->mu);
+    std::lock_guard<std::recursive_mutex> g(c->mu);
     c->n_launches = 0;
     c->kernel_ms = 0.0;
     c->bytes_algo = 0;
@@ -1240,7 +1297,7 @@ extern "C" int ua_stats_reset(ua_ctx *c) {
 
 extern "C" int ua_stats_get(ua_ctx *c, uint64_t *n_launches, double *kernel_ms,
                             uint64_t *bytes_algorithmic) {
-    std::lock_guard<std::mutex> g(c->mu);
+    std::lock_guard<std::recursive_mutex> g(c->mu);
     if (n_launches) *n_launches = c->n_launches;
     if (kernel_ms) *kernel_ms = c->kernel_ms;
     if (bytes_algorithmic) *bytes_algorithmic = c->bytes_algo;
@@ -1254,13 +1311,15 @@ static void launch_tiles(ua_ctx *c, const UaDesc *descs, const u32 *tpair,
                          const u32 *ta0, u64 T, u64 *stage, u64 stride, u32 *tcnt,
                          const u64 *offs, const u64 *part) {
 #if UA_PIPE
-    hipLaunchKernelGGL((k_tiles_pipe<OP, MODE>), dim3((u32)((T + 1) / 2)),
-                       dim3(UA_BLOCK), 0, c->stream, descs, tpair, ta0, T, stage,
-                       stride, tcnt, offs, part);
-#else
+    if constexpr (MODE != MODE_DIRECT) {
+        hipLaunchKernelGGL((k_tiles_pipe<OP, MODE>), dim3((u32)((T + 1) / 2)),
+                           dim3(UA_BLOCK), 0, c->stream, descs, tpair, ta0, T, stage,
+                           stride, tcnt, offs, part);
+        return;
+    }
+#endif
     hipLaunchKernelGGL((k_tiles<OP, MODE>), dim3((u32)T), dim3(UA_BLOCK), 0, c->stream,
                        descs, tpair, ta0, T, stage, stride, tcnt, offs, part);
-#endif
 }
 
 /* Split flat scan: offs_dev gets chunk-local exclusive offsets, WS_PARTIAL
@@ -1337,6 +1396,25 @@ static int run_batch_locked(ua_ctx *c, const ua_dpair *pairs, int n_pairs,
         hipLaunchKernelGGL(k_partition, dim3((u32)pblk), dim3(UA_BLOCK), 0, c->stream,
                            d_descs, d_tb, n_pairs, total_tiles, d_tpair, d_ta0, 1);
 
+        if (op == OP_MERGE_ALL) {
+            /* duplicate-keeping merge: output position == path position, so no
+             * scan / staging / compaction / pair_out — lens are n+m */
+            HIP_TRY(hipEventRecord(c->ev[0], c->stream));
+            launch_tiles<OP_MERGE_ALL, MODE_DIRECT>(c, d_descs, d_tpair, d_ta0,
+                                                    total_tiles, nullptr, 0, d_tcnt,
+                                                    nullptr, nullptr);
+            HIP_TRY(hipEventRecord(c->ev[1], c->stream));
+            HIP_TRY(hipStreamSynchronize(c->stream));
+            HIP_TRY(hipGetLastError());
+            float ms = 0.f;
+            HIP_TRY(hipEventElapsedTime(&ms, c->ev[0], c->ev[1]));
+            c->kernel_ms += ms;
+            c->n_launches += 1;
+            for (int p = 0; p < n_pairs; p++) out_lens[p] = pairs[p].n + pairs[p].m;
+            c->bytes_algo += 2 * in_bytes;
+            return UA_OK;
+        }
+
         HIP_TRY(hipEventRecord(c->ev[0], c->stream));
         if (op == OP_INTERSECT) {
             launch_tiles<OP_INTERSECT, MODE_STAGE>(c, d_descs, d_tpair, d_ta0,
@@ -1400,7 +1478,7 @@ static int run_batch_locked(ua_ctx *c, const ua_dpair *pairs, int n_pairs,
 }
 
 static int run_batch(ua_ctx *c, const ua_dpair *pairs, int n_pairs, uint64_t *out_lens, int op) {
-    std::lock_guard<std::mutex> g(c->mu);
+    std::lock_guard<std::recursive_mutex> g(c->mu);
     return run_batch_locked(c, pairs, n_pairs, out_lens, op);
 }
 
@@ -1432,7 +1510,7 @@ static size_t align16(size_t x) { return (x + 15) & ~(size_t)15; }
 
 extern "C" int ua_batch_create(ua_ctx *c, const ua_dpair *pairs, int n_pairs,
                                ua_batch **out) {
-    std::lock_guard<std::mutex> g(c->mu);
+    std::lock_guard<std::recursive_mutex> g(c->mu);
     HIP_TRY(hipSetDevice(c->device));
     ua_batch *b = new ua_batch();
     b->n_pairs = n_pairs;
@@ -1510,7 +1588,7 @@ extern "C" void ua_batch_destroy(ua_ctx *c, ua_batch *b) {
 }
 
 extern "C" int ua_batch_run(ua_ctx *c, ua_batch *b, int op, uint64_t *out_lens) {
-    std::lock_guard<std::mutex> g(c->mu);
+    std::lock_guard<std::recursive_mutex> g(c->mu);
     HIP_TRY(hipSetDevice(c->device));
     if (b->n_pairs == 0) return UA_OK;
     int kop = (op == UA_OP_INTERSECT) ? OP_INTERSECT
@@ -1595,9 +1673,81 @@ extern "C" int ua_difference_batch_dev(ua_ctx *c, const ua_dpair *pairs, int n_p
     return run_batch(c, pairs, n_pairs, out_lens, OP_DIFF);
 }
 
+/* duplicate-keeping batched merge (sorted-run merge; out capacity exactly n+m) */
+extern "C" int ua_merge_all_batch_dev(ua_ctx *c, const ua_dpair *pairs, int n_pairs,
+                                      uint64_t *out_lens) {
+    return run_batch(c, pairs, n_pairs, out_lens, OP_MERGE_ALL);
+}
+
+/* ---- batched segmented sort: LDS bitonic chunks + OP_MERGE_ALL tree ---- */
+extern "C" int ua_sort_segments_dev(ua_ctx *c, const ua_dseg *segs, int n_segs) {
+    std::lock_guard<std::recursive_mutex> gop(c->mu);
+    HIP_TRY(hipSetDevice(c->device));
+    if (n_segs <= 0) return UA_OK;
+
+    u64 max_runs = 1;
+    std::vector<UaChunk> chunks;
+    for (int sgi = 0; sgi < n_segs; sgi++) {
+        u64 n = segs[sgi].n;
+        u64 runs = (n + UA_SORT_N - 1) / UA_SORT_N;
+        if (runs > max_runs) max_runs = runs;
+    }
+    int rounds = 0;
+    for (u64 w = 1; w < max_runs; w <<= 1) rounds++;
+    /* choose the chunk-stage output side so the final merge lands in data */
+    int start_in_tmp = rounds & 1;
+
+    for (int sgi = 0; sgi < n_segs; sgi++) {
+        const ua_dseg &sg = segs[sgi];
+        u64 *dst = start_in_tmp ? sg.tmp : sg.data;
+        for (u64 off = 0; off < sg.n; off += UA_SORT_N) {
+            u32 len = (u32)std::min<u64>(UA_SORT_N, sg.n - off);
+            chunks.push_back({sg.data + off, dst + off, len, 0});
+        }
+    }
+    if (!chunks.empty()) {
+        int rc;
+        if ((rc = ws_reserve(c, WS_DESC, chunks.size() * sizeof(UaChunk)))) return rc;
+        HIP_TRY(hipMemcpyAsync(c->ws[WS_DESC], chunks.data(),
+                               chunks.size() * sizeof(UaChunk), hipMemcpyHostToDevice,
+                               c->stream));
+        hipLaunchKernelGGL(k_sort_chunks, dim3((u32)chunks.size()), dim3(UA_BLOCK), 0,
+                           c->stream, (const UaChunk *)c->ws[WS_DESC],
+                           (u64)chunks.size());
+        HIP_TRY(hipStreamSynchronize(c->stream));
+    }
+
+    u64 width = UA_SORT_N;
+    int side = start_in_tmp; /* 1 = current sorted runs live in tmp */
+    for (int r = 0; r < rounds; r++) {
+        std::vector<ua_dpair> prs;
+        for (int sgi = 0; sgi < n_segs; sgi++) {
+            const ua_dseg &sg = segs[sgi];
+            u64 *cur = side ? sg.tmp : sg.data;
+            u64 *nxt = side ? sg.data : sg.tmp;
+            for (u64 off = 0; off < sg.n; off += 2 * width) {
+                u64 n1 = std::min<u64>(width, sg.n - off);
+                u64 rem = sg.n - off - n1;
+                u64 n2 = std::min<u64>(width, rem);
+                /* odd trailing run pairs with an empty side = a copy */
+                prs.push_back({cur + off, n1, cur + off + n1, n2, nxt + off});
+            }
+        }
+        if (!prs.empty()) {
+            std::vector<u64> lens(prs.size());
+            int rc = run_batch_locked(c, prs.data(), (int)prs.size(), lens.data(),
+                                      OP_MERGE_ALL);
+            if (rc) return rc;
+        }
+        width <<= 1;
+        side ^= 1;
+    }
+    return UA_OK; /* side == 0: result in data */
+}
+
 extern "C" int ua_index_of_batch_dev(ua_ctx *c, const uint64_t *u, uint64_t n,
                                      const uint64_t *queries, uint64_t nq, int64_t *out) {
-    std::lock_guard<std::mutex> g(c->mu);
+    std::lock_guard<std::recursive_mutex> g(c->mu);
     HIP_TRY(hipSetDevice(c->device));
     if (nq == 0) return UA_OK;
     u64 nblk = (nq + UA_BLOCK - 1) / UA_BLOCK;
@@ -1612,6 +1762,7 @@ extern "C" int ua_index_of_batch_dev(ua_ctx *c, const uint64_t *u, uint64_t n,
 extern "C" int ua_intersect_k_dev(ua_ctx *c, const uint64_t *const *lists,
                                   const uint64_t *lens, int k, uint64_t *out,
                                   uint64_t *out_n) {
+    std::lock_guard<std::recursive_mutex> gop(c->mu); /* whole-op: scratch reuse */
     if (k <= 0) {
         *out_n = 0;
         return UA_OK;
@@ -1620,7 +1771,7 @@ extern "C" int ua_intersect_k_dev(ua_ctx *c, const uint64_t *const *lists,
     for (int i = 0; i < k; i++) ord[i] = i;
     std::stable_sort(ord.begin(), ord.end(), [&](int a, int b) { return lens[a] < lens[b]; });
     if (k == 1) {
-        std::lock_guard<std::mutex> g(c->mu);
+        std::lock_guard<std::recursive_mutex> g(c->mu);
         HIP_TRY(hipSetDevice(c->device));
         HIP_TRY(hipMemcpyAsync(out, lists[0], lens[0] * sizeof(u64),
                                hipMemcpyDeviceToDevice, c->stream));
@@ -1631,7 +1782,7 @@ extern "C" int ua_intersect_k_dev(ua_ctx *c, const uint64_t *const *lists,
     u64 cap = lens[ord[0]];
     int rc;
     {
-        std::lock_guard<std::mutex> g(c->mu);
+        std::lock_guard<std::recursive_mutex> g(c->mu);
         if ((rc = ws_reserve(c, WS_SCRATCH_A, (cap ? cap : 1) * sizeof(u64)))) return rc;
     }
     u64 *scratch = (u64 *)c->ws[WS_SCRATCH_A];
@@ -1642,7 +1793,7 @@ extern "C" int ua_intersect_k_dev(ua_ctx *c, const uint64_t *const *lists,
     for (int j = 2; j < k && len0 > 0; j++) {
         pr = {out, len0, lists[ord[j]], lens[ord[j]], scratch};
         if ((rc = run_batch(c, &pr, 1, &len0, OP_INTERSECT))) return rc;
-        std::lock_guard<std::mutex> g(c->mu);
+        std::lock_guard<std::recursive_mutex> g(c->mu);
         HIP_TRY(hipSetDevice(c->device));
         HIP_TRY(hipMemcpyAsync(out, scratch, len0 * sizeof(u64),
                                hipMemcpyDeviceToDevice, c->stream));
@@ -1656,6 +1807,7 @@ extern "C" int ua_intersect_k_dev(ua_ctx *c, const uint64_t *const *lists,
 extern "C" int ua_merge_k_dev(ua_ctx *c, const uint64_t *const *lists,
                               const uint64_t *lens, int k, uint64_t *out,
                               uint64_t *out_n) {
+    std::lock_guard<std::recursive_mutex> gop(c->mu); /* whole-op: scratch reuse */
     if (k <= 0) {
         *out_n = 0;
         return UA_OK;
@@ -1664,7 +1816,7 @@ extern "C" int ua_merge_k_dev(ua_ctx *c, const uint64_t *const *lists,
     for (int i = 0; i < k; i++) total += lens[i];
     int rc;
     {
-        std::lock_guard<std::mutex> g(c->mu);
+        std::lock_guard<std::recursive_mutex> g(c->mu);
         if ((rc = ws_reserve(c, WS_SCRATCH_A, (total ? total : 1) * sizeof(u64)))) return rc;
         if ((rc = ws_reserve(c, WS_SCRATCH_B, (total ? total : 1) * sizeof(u64)))) return rc;
     }
@@ -1702,7 +1854,7 @@ extern "C" int ua_merge_k_dev(ua_ctx *c, const uint64_t *const *lists,
         which ^= 1;
     }
     {
-        std::lock_guard<std::mutex> g(c->mu);
+        std::lock_guard<std::recursive_mutex> g(c->mu);
         HIP_TRY(hipSetDevice(c->device));
         HIP_TRY(hipMemcpyAsync(out, cur_ptr[0], cur_len[0] * sizeof(u64),
                                hipMemcpyDeviceToDevice, c->stream));
@@ -1776,7 +1928,7 @@ static int run_packed_locked(ua_ctx *c, const ua_dpack *pk, u64 after, const u64
 extern "C" int ua_intersect_packed_dev(ua_ctx *c, const ua_dpack *pk, uint64_t after_uid,
                                        const uint64_t *v, uint64_t m, uint64_t *out,
                                        uint64_t *out_n) {
-    std::lock_guard<std::mutex> g(c->mu);
+    std::lock_guard<std::recursive_mutex> g(c->mu);
     return run_packed_locked(c, pk, after_uid, v, m, out, out_n, 0);
 }
 
@@ -1788,7 +1940,7 @@ extern "C" int ua_intersect_packed_batch_dev(ua_ctx *c, const uint64_t *bases,
                                              const uint64_t *pack_block_base,
                                              int n_packs, const ua_ptask *tasks,
                                              uint64_t *out_lens) {
-    std::lock_guard<std::mutex> g(c->mu);
+    std::lock_guard<std::recursive_mutex> g(c->mu);
     HIP_TRY(hipSetDevice(c->device));
     if (n_packs <= 0) return UA_OK;
     u64 nb = pack_block_base[n_packs];
@@ -1860,7 +2012,7 @@ extern "C" int ua_intersect_packed_batch_dev(ua_ctx *c, const uint64_t *bases,
 
 extern "C" int ua_decode_dev(ua_ctx *c, const ua_dpack *pk, uint64_t seek_uid, uint64_t *out,
                              uint64_t *out_n) {
-    std::lock_guard<std::mutex> g(c->mu);
+    std::lock_guard<std::recursive_mutex> g(c->mu);
     return run_packed_locked(c, pk, seek_uid, nullptr, 0, out, out_n, 1);
 }
 
@@ -1869,7 +2021,7 @@ extern "C" int ua_encode_dev(ua_ctx *c, const uint64_t *uids, uint64_t n,
                              uint32_t block_size, uint64_t *bases, uint32_t *num_uids,
                              uint64_t *delta_offs, uint8_t *deltas,
                              uint64_t *n_blocks_out, uint64_t *deltas_bytes_out) {
-    std::lock_guard<std::mutex> g(c->mu);
+    std::lock_guard<std::recursive_mutex> g(c->mu);
     HIP_TRY(hipSetDevice(c->device));
     if (block_size > UA_MAX_BLOCK_UIDS) return UA_ERR_INVALID;
     if (n == 0) {
@@ -2082,9 +2234,10 @@ extern "C" int64_t ua_index_of(const uint64_t *u, uint64_t n, uint64_t uid) {
 
 static int host_pair_op(ua_ctx *c, const u64 *u, u64 n, const u64 *v, u64 m, u64 *out,
                         u64 *out_n, int op) {
+    std::lock_guard<std::recursive_mutex> gop(c->mu); /* whole-op: WS_H* reuse */
     int rc;
     {
-        std::lock_guard<std::mutex> g(c->mu);
+        std::lock_guard<std::recursive_mutex> g(c->mu);
         HIP_TRY(hipSetDevice(c->device));
         if ((rc = ws_reserve(c, WS_HU, (n ? n : 1) * sizeof(u64)))) return rc;
         if ((rc = ws_reserve(c, WS_HV, (m ? m : 1) * sizeof(u64)))) return rc;
@@ -2100,7 +2253,7 @@ static int host_pair_op(ua_ctx *c, const u64 *u, u64 n, const u64 *v, u64 m, u64
     u64 len = 0;
     if ((rc = run_batch(c, &pr, 1, &len, op))) return rc;
     {
-        std::lock_guard<std::mutex> g(c->mu);
+        std::lock_guard<std::recursive_mutex> g(c->mu);
         HIP_TRY(hipMemcpyAsync(out, c->ws[WS_HOUT], len * sizeof(u64),
                                hipMemcpyDeviceToHost, c->stream));
         HIP_TRY(hipStreamSynchronize(c->stream));
@@ -2140,6 +2293,7 @@ static int host_upload_lists(ua_ctx *c, const uint64_t *const *lists, const uint
 
 extern "C" int ua_intersect_k(ua_ctx *c, const uint64_t *const *lists, const uint64_t *lens,
                               int k, uint64_t *out, uint64_t *out_n) {
+    std::lock_guard<std::recursive_mutex> gop(c->mu);
     if (k <= 0) {
         *out_n = 0;
         return UA_OK;
@@ -2148,7 +2302,7 @@ extern "C" int ua_intersect_k(ua_ctx *c, const uint64_t *const *lists, const uin
     u64 total;
     int rc;
     {
-        std::lock_guard<std::mutex> g(c->mu);
+        std::lock_guard<std::recursive_mutex> g(c->mu);
         HIP_TRY(hipSetDevice(c->device));
         if ((rc = host_upload_lists(c, lists, lens, k, dptrs, total))) return rc;
         u64 cap = lens[0];
@@ -2159,7 +2313,7 @@ extern "C" int ua_intersect_k(ua_ctx *c, const uint64_t *const *lists, const uin
     if ((rc = ua_intersect_k_dev(c, dptrs.data(), lens, k, (u64 *)c->ws[WS_HOUT], &len)))
         return rc;
     {
-        std::lock_guard<std::mutex> g(c->mu);
+        std::lock_guard<std::recursive_mutex> g(c->mu);
         HIP_TRY(hipMemcpyAsync(out, c->ws[WS_HOUT], len * sizeof(u64),
                                hipMemcpyDeviceToHost, c->stream));
         HIP_TRY(hipStreamSynchronize(c->stream));
@@ -2170,6 +2324,7 @@ extern "C" int ua_intersect_k(ua_ctx *c, const uint64_t *const *lists, const uin
 
 extern "C" int ua_merge_k(ua_ctx *c, const uint64_t *const *lists, const uint64_t *lens,
                           int k, uint64_t *out, uint64_t *out_n) {
+    std::lock_guard<std::recursive_mutex> gop(c->mu);
     if (k <= 0) {
         *out_n = 0;
         return UA_OK;
@@ -2178,7 +2333,7 @@ extern "C" int ua_merge_k(ua_ctx *c, const uint64_t *const *lists, const uint64_
     u64 total;
     int rc;
     {
-        std::lock_guard<std::mutex> g(c->mu);
+        std::lock_guard<std::recursive_mutex> g(c->mu);
         HIP_TRY(hipSetDevice(c->device));
         if ((rc = host_upload_lists(c, lists, lens, k, dptrs, total))) return rc;
         if ((rc = ws_reserve(c, WS_HOUT, (total ? total : 1) * sizeof(u64)))) return rc;
@@ -2187,7 +2342,7 @@ extern "C" int ua_merge_k(ua_ctx *c, const uint64_t *const *lists, const uint64_
     if ((rc = ua_merge_k_dev(c, dptrs.data(), lens, k, (u64 *)c->ws[WS_HOUT], &len)))
         return rc;
     {
-        std::lock_guard<std::mutex> g(c->mu);
+        std::lock_guard<std::recursive_mutex> g(c->mu);
         HIP_TRY(hipMemcpyAsync(out, c->ws[WS_HOUT], len * sizeof(u64),
                                hipMemcpyDeviceToHost, c->stream));
         HIP_TRY(hipStreamSynchronize(c->stream));
@@ -2199,6 +2354,7 @@ extern "C" int ua_merge_k(ua_ctx *c, const uint64_t *const *lists, const uint64_
 extern "C" int ua_intersect_packed(ua_ctx *c, const ua_pack *pack, uint64_t after_uid,
                                    const uint64_t *v, uint64_t m, uint64_t *out,
                                    uint64_t *out_n) {
+    std::lock_guard<std::recursive_mutex> gop(c->mu);
     if (!pack || pack->n_blocks == 0) {
         *out_n = 0;
         return UA_OK;
@@ -2213,7 +2369,7 @@ extern "C" int ua_intersect_packed(ua_ctx *c, const ua_pack *pack, uint64_t afte
         return rc;
     ua_dpack dpk;
     {
-        std::lock_guard<std::mutex> g(c->mu);
+        std::lock_guard<std::recursive_mutex> g(c->mu);
         HIP_TRY(hipSetDevice(c->device));
         size_t need = nb * 8 + (nb + 1) * 8 + nb * 4 + blob.size() + 64;
         if ((rc = ws_reserve(c, WS_PACK, need))) return rc;
@@ -2241,7 +2397,7 @@ extern "C" int ua_intersect_packed(ua_ctx *c, const ua_pack *pack, uint64_t afte
                                       (u64 *)c->ws[WS_HOUT], &len)))
         return rc;
     {
-        std::lock_guard<std::mutex> g(c->mu);
+        std::lock_guard<std::recursive_mutex> g(c->mu);
         HIP_TRY(hipMemcpyAsync(out, c->ws[WS_HOUT], len * sizeof(u64),
                                hipMemcpyDeviceToHost, c->stream));
         HIP_TRY(hipStreamSynchronize(c->stream));

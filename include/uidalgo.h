@@ -152,6 +152,23 @@ int ua_merge_batch_dev(ua_ctx *, const ua_dpair *pairs, int n_pairs,
 int ua_difference_batch_dev(ua_ctx *, const ua_dpair *pairs, int n_pairs,
                             uint64_t *out_lens);
 
+/* batched duplicate-KEEPING merge of sorted runs (no dedup, unlike
+ * MergeSorted): out capacity exactly n+m; out_lens[i] = n+m.  The building
+ * block of the segmented sort. */
+int ua_merge_all_batch_dev(ua_ctx *, const ua_dpair *pairs, int n_pairs,
+                           uint64_t *out_lens);
+
+/* Batched segmented sort, u64 ascending, duplicates kept, in place.
+ * tmp: device scratch with capacity n per segment.  The engine primitive
+ * behind the sort path's UidMatrix shapes (worker/sort.go:48,139,189 —
+ * SURVEY.md §8f row 3) and for unsorted ingest ahead of Encode. */
+typedef struct {
+    uint64_t *data; /* device, n elements; sorted ascending on return */
+    uint64_t n;
+    uint64_t *tmp;  /* device scratch, capacity n */
+} ua_dseg;
+int ua_sort_segments_dev(ua_ctx *, const ua_dseg *segs, int n_segs);
+
 /* k-way fold algo.IntersectSorted (uidlist.go:297): lists/lens host arrays of
  * device pointers; out device, capacity min(lens). */
 int ua_intersect_k_dev(ua_ctx *, const uint64_t *const *lists,

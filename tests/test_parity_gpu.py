@@ -226,6 +226,45 @@ def test_kway_vs_oracle(eng, k):
     assert got_i.tolist() == orc.intersect_sorted(lists).tolist()
 
 
+def test_merge_all_pairs(eng):
+    """Duplicate-keeping merge of sorted runs (the segmented-sort building
+    block): equals np.sort of the concatenation."""
+    rng = np.random.default_rng(SEED + 5)
+    us, vs = [], []
+    for _ in range(16):
+        n, m = int(rng.integers(0, 30_000)), int(rng.integers(0, 30_000))
+        us.append(np.sort(rng.integers(0, 40_000, size=n, dtype=np.uint64)))
+        vs.append(np.sort(rng.integers(0, 40_000, size=m, dtype=np.uint64)))
+    outs, lens = eng.merge_all_pairs([to_dev(u) for u in us], [to_dev(v) for v in vs])
+    for i in range(16):
+        assert lens[i] == us[i].size + vs[i].size
+        got = to_np(outs[i][:lens[i]])
+        want = np.sort(np.concatenate([us[i], vs[i]]))
+        assert np.array_equal(got, want), f"pair {i}"
+
+
+@pytest.mark.parametrize("sizes", [
+    [0, 1, 5, 2047, 2048, 2049],
+    [100_000, 3, 1_000_000],
+    [65_536] * 8,
+])
+def test_sort_segments(eng, sizes):
+    """Segmented sort (bitonic chunks + merge-all tree) vs np.sort — with
+    duplicates, u64-extreme values, in place."""
+    rng = np.random.default_rng(SEED + sum(sizes))
+    arrs = []
+    for n in sizes:
+        a = rng.integers(0, max(n, 10) * 2, size=n, dtype=np.uint64)
+        if n > 10:  # sprinkle extremes and duplicates
+            a[:: max(n // 7, 1)] = np.uint64(2**64 - 1)
+            a[1:: max(n // 5, 1)] = a[0]
+        arrs.append(a)
+    tensors = [to_dev(a) for a in arrs]
+    eng.sort_segments(tensors)
+    for a, t in zip(arrs, tensors):
+        assert np.array_equal(to_np(t), np.sort(a))
+
+
 def test_index_of_batch_vs_oracle(eng):
     rng = np.random.default_rng(SEED)
     u = synth.gen_sorted_unique(rng, 100_000, 1_000_000)
