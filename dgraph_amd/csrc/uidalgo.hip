@@ -55,6 +55,12 @@ typedef uint8_t u8;
 #define UA_MAX_BLOCK_UIDS 256
 #define UA_MAX_DELTAS 1092 /* 64 groups x 17 B, padded */
 
+static_assert(UA_TILE % UA_BLOCK == 0, "tile must divide evenly over threads");
+static_assert(UA_TILE / UA_BLOCK >= 1 && UA_TILE / UA_BLOCK <= 32,
+              "emission flags are one u32 bit per path step per thread");
+static_assert(UA_BLOCK == 256, "wave geometry (4 waves/WG) is hard-coded in "
+                               "scan/compact/packed kernels");
+
 /* ==================== device helpers ==================== */
 
 __device__ __forceinline__ u64 d_lower_bound(const u64 *__restrict__ a, u64 n, u64 key) {
