@@ -1693,19 +1693,23 @@ extern "C" int ua_sort_segments_dev(ua_ctx *c, const ua_dseg *segs, int n_segs) 
 
     u64 max_runs = 1;
     std::vector<UaChunk> chunks;
+    std::vector<int> seg_rounds(n_segs);
     for (int sgi = 0; sgi < n_segs; sgi++) {
         u64 n = segs[sgi].n;
         u64 runs = (n + UA_SORT_N - 1) / UA_SORT_N;
         if (runs > max_runs) max_runs = runs;
+        int r = 0;
+        for (u64 w = 1; w < runs; w <<= 1) r++;
+        seg_rounds[sgi] = r;
     }
     int rounds = 0;
     for (u64 w = 1; w < max_runs; w <<= 1) rounds++;
-    /* choose the chunk-stage output side so the final merge lands in data */
-    int start_in_tmp = rounds & 1;
-
+    /* per-segment start side: a segment with r merge rounds of its own flips
+     * sides r times, so start it where it will END in data — finished
+     * segments then never re-copy while bigger ones keep merging */
     for (int sgi = 0; sgi < n_segs; sgi++) {
         const ua_dseg &sg = segs[sgi];
-        u64 *dst = start_in_tmp ? sg.tmp : sg.data;
+        u64 *dst = (seg_rounds[sgi] & 1) ? sg.tmp : sg.data;
         for (u64 off = 0; off < sg.n; off += UA_SORT_N) {
             u32 len = (u32)std::min<u64>(UA_SORT_N, sg.n - off);
             chunks.push_back({sg.data + off, dst + off, len, 0});
@@ -1724,11 +1728,13 @@ extern "C" int ua_sort_segments_dev(ua_ctx *c, const ua_dseg *segs, int n_segs) 
     }
 
     u64 width = UA_SORT_N;
-    int side = start_in_tmp; /* 1 = current sorted runs live in tmp */
     for (int r = 0; r < rounds; r++) {
         std::vector<ua_dpair> prs;
         for (int sgi = 0; sgi < n_segs; sgi++) {
+            if (r >= seg_rounds[sgi]) continue; /* segment already fully sorted */
             const ua_dseg &sg = segs[sgi];
+            /* this segment has flipped r times from its start side */
+            int side = (seg_rounds[sgi] - r) & 1; /* 1 = current runs in tmp */
             u64 *cur = side ? sg.tmp : sg.data;
             u64 *nxt = side ? sg.data : sg.tmp;
             for (u64 off = 0; off < sg.n; off += 2 * width) {
@@ -1746,9 +1752,8 @@ extern "C" int ua_sort_segments_dev(ua_ctx *c, const ua_dseg *segs, int n_segs) 
             if (rc) return rc;
         }
         width <<= 1;
-        side ^= 1;
     }
-    return UA_OK; /* side == 0: result in data */
+    return UA_OK; /* every segment ends in data by start-side parity */
 }
 
 extern "C" int ua_index_of_batch_dev(ua_ctx *c, const uint64_t *u, uint64_t n,
