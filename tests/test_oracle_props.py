@@ -111,6 +111,50 @@ def test_decoder_seek_decode_suffix():
         assert pack.decode(i).tolist() == uids[start:].tolist()
 
 
+def test_seek_to_block_semantics():
+    """codec.Decoder.SeekToBlock (codec.go:219): positions at the block that
+    could contain uid and returns the WHOLE block untruncated (unlike Seek);
+    stateful prevBlockIdx fast path included."""
+    uids = np.arange(0, 10001, 10, dtype=np.uint64)
+    pack = orc.Pack(uids, 10)
+    bases, nums, offs, blob = pack.flatten()
+    dec = orc.Dec(pack)
+    dec.seek(0, orc.SEEK_START)
+    for x in range(5, 10000, 97):
+        got = dec.seek_to_block(x, orc.SEEK_CURRENT)
+        # whence=SeekCurrent: last block with base <= x; if x exceeds that
+        # block's last uid, SeekToBlock falls through to Next() (codec.go:270)
+        bidx = max(int(np.searchsorted(bases, x, side="right")) - 1, 0)
+        last_uid = int(bases[bidx]) + (int(nums[bidx]) - 1) * 10
+        if x > last_uid:
+            bidx += 1
+        assert got.size > 0 and got[0] == bases[bidx], (x, got[:3])
+        # whole block, no truncation to >= x
+        assert got.size == nums[bidx]
+    # exhausted decoder (past the end) returns empty where Go would panic
+    dec2 = orc.Dec(pack)
+    dec2.seek(10**9, orc.SEEK_START)
+    got = dec2.seek_to_block(50, orc.SEEK_CURRENT)
+    assert got.size == 0
+
+
+def test_decoder_valid_and_next():
+    uids = np.arange(0, 100, 3, dtype=np.uint64)
+    pack = orc.Pack(uids, 10)
+    dec = orc.Dec(pack)
+    dec.seek(0, orc.SEEK_START)
+    seen = []
+    while True:
+        u = dec.uids()
+        if u.size == 0:
+            break
+        seen.extend(u.tolist())
+        if not dec.valid():
+            break
+        dec.next()
+    assert seen == uids.tolist()
+
+
 def fill_nums(rng, n1, n2):
     """Mirrors uidlist_test.go:583-605 fillNums: returns (common, block, other)."""
     common = rng.integers(0, 2**64, size=n1, dtype=np.uint64)
