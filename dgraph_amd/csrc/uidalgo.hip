@@ -188,9 +188,29 @@ enum { MODE_STAGE = 0, MODE_COUNT = 1, MODE_WRITE = 2, MODE_DIRECT = 3 };
  * ONCE: count + payload together, no re-walk, no scratch spill (guide §5.4
  * rule 20: runtime-indexed local arrays go to scratch; per-step static
  * indices do not). */
+/* UNION dedup: the reference's MergeSorted emits the merged stream with ALL
+ * duplicates collapsed (uidlist.go:417: output==last skips), including
+ * duplicates WITHIN one list (TestMergeSorted6/9/10).  Rule: emit iff value
+ * != previous emitted value == previous merged-stream value.  A thread's
+ * initial "previous" is max(last A before it, last B before it), using the
+ * staged tile-boundary elements when its segment starts at a range head. */
+__device__ __forceinline__ u64 d_prev_stream(const u64 *As, int i0, u64 a_before,
+                                             bool has_ab, const u64 *Bs, int j0,
+                                             u64 b_before, bool has_bb, bool &has_prev) {
+    u64 pa = (i0 > 0) ? As[i0 - 1] : a_before;
+    bool hpa = (i0 > 0) || has_ab;
+    u64 pb = (j0 > 0) ? Bs[j0 - 1] : b_before;
+    bool hpb = (j0 > 0) || has_bb;
+    has_prev = hpa || hpb;
+    if (!hpa) return pb;
+    if (!hpb) return pa;
+    return pa > pb ? pa : pb;
+}
+
 template <int OP>
 __device__ __forceinline__ int tile_walk(const u64 *As, int alen, const u64 *Bs, int blen,
-                                         u64 a_before, bool has_ab, bool has_bn,
+                                         u64 a_before, bool has_ab, u64 b_before,
+                                         bool has_bb, bool has_bn,
                                          int s0, int s1, int i0,
                                          u64 (&em)[UA_WPT], u32 &flags) {
     int i = i0, j = s0 - i0;
@@ -202,7 +222,11 @@ __device__ __forceinline__ int tile_walk(const u64 *As, int alen, const u64 *Bs,
      * instead of re-gathering As[i]/Bs[j] every comparison */
     u64 a = (i < alen) ? As[i] : 0;
     u64 b = (j < blen_ext) ? Bs[j] : 0;
-    u64 prev_a = (i > 0) ? As[i - 1] : a_before; /* union dedup lookback */
+    bool has_prev = false;
+    u64 prev_out = 0;
+    if (OP == OP_UNION)
+        prev_out = d_prev_stream(As, i, a_before, has_ab, Bs, j, b_before, has_bb,
+                                 has_prev);
 #pragma unroll
     for (int s = 0; s < UA_WPT; s++) {
         if (s >= steps || (i >= alen && j >= blen)) break;
@@ -222,22 +246,26 @@ __device__ __forceinline__ int tile_walk(const u64 *As, int alen, const u64 *Bs,
                     flags |= 1u << s;
                     cnt++;
                 }
-            } else { /* UNION: A always emits */
-                em[s] = a;
-                flags |= 1u << s;
-                cnt++;
+            } else { /* UNION: emit iff != previous merged-stream value */
+                if (!has_prev || a != prev_out) {
+                    em[s] = a;
+                    flags |= 1u << s;
+                    cnt++;
+                }
+                prev_out = a;
+                has_prev = true;
             }
-            if (OP == OP_UNION) prev_a = a;
             i++;
             if (i < alen) a = As[i];
         } else {
             if (OP == OP_UNION) {
-                bool dup = (i > 0 || has_ab) && (b == prev_a);
-                if (!dup) {
+                if (!has_prev || b != prev_out) {
                     em[s] = b;
                     flags |= 1u << s;
                     cnt++;
                 }
+                prev_out = b;
+                has_prev = true;
             }
             j++;
             if (j < blen_ext) b = Bs[j];
@@ -328,6 +356,7 @@ __device__ __forceinline__ int tile_search(const u64 *As, int alen, const u64 *B
 template <int OP>
 __device__ __forceinline__ int tile_walk2(const u64 *smembase, int aoff, int alen,
                                           int boff, int blen, u64 a_before, bool has_ab,
+                                          u64 b_before, bool has_bb,
                                           bool has_bn, int s0, int s1, int i0,
                                           u64 (&em)[UA_WPT], u32 &flags) {
     const u64 *As = smembase + aoff;
@@ -345,7 +374,11 @@ __device__ __forceinline__ int tile_walk2(const u64 *smembase, int aoff, int ale
     u64 an = As[(i + 1) < alen ? (i + 1) : amax];
     u64 b = Bs[j < blen_ext ? j : bmax];
     u64 bn = Bs[(j + 1) < blen_ext ? (j + 1) : bmax];
-    u64 prev_a = (i > 0) ? As[i - 1] : a_before;
+    bool has_prev = false;
+    u64 prev_out = 0;
+    if (OP == OP_UNION)
+        prev_out = d_prev_stream(As, i, a_before, has_ab, Bs, j, b_before, has_bb,
+                                 has_prev);
 #pragma unroll
     for (int s = 0; s < UA_WPT; s++) {
         if (s >= steps) break;
@@ -363,11 +396,12 @@ __device__ __forceinline__ int tile_walk2(const u64 *smembase, int aoff, int ale
         } else if (OP == OP_MERGE_ALL) { /* duplicate-keeping merge (sort runs) */
             emit = true;
             em[s] = takeA ? a : b;
-        } else { /* UNION */
-            bool dupB = (i > 0 || has_ab) && (b == prev_a);
-            emit = takeA || !dupB;
-            em[s] = takeA ? a : b;
-            prev_a = takeA ? a : prev_a;
+        } else { /* UNION: emit iff != previous merged-stream value */
+            u64 val = takeA ? a : b;
+            emit = !has_prev || val != prev_out;
+            em[s] = val;
+            prev_out = val;
+            has_prev = true;
         }
         flags |= ((u32)emit) << s;
         cnt += emit;
@@ -419,6 +453,7 @@ __global__ __launch_bounds__(UA_BLOCK) void k_tiles(
     __shared__ __align__(16) u64 smem[UA_TILE + 4];
     __shared__ u32 scan[UA_BLOCK / 64]; /* per-wave totals for d_block_scan */
     __shared__ u64 s_abefore;
+    __shared__ u64 s_bbefore;
 
     u64 t = blockIdx.x;
     int tid = threadIdx.x;
@@ -438,12 +473,14 @@ __global__ __launch_bounds__(UA_BLOCK) void k_tiles(
     u64 *Bs = smem + ((alen + 1) & ~1);      /* rounded up to even: 16-B aligned */
 
     bool has_ab = (a0 > 0);
+    bool has_bb = (b0 > 0);
     bool has_bn = ((u64)b1 < d.m);
 #if UA_ABLATE != 2
     d_fill_lds(As, d.u + a0, alen, tid);
     d_fill_lds(Bs, d.v + b0, blen, tid);
     if (tid == 0) {
         s_abefore = has_ab ? d.u[a0 - 1] : 0;
+        s_bbefore = has_bb ? d.v[b0 - 1] : 0;
         Bs[blen] = has_bn ? d.v[b1] : 0;
     }
 #endif
@@ -476,12 +513,14 @@ __global__ __launch_bounds__(UA_BLOCK) void k_tiles(
         if (s1 > tilelen) s1 = tilelen;
         int i0 = d_merge_path_lds(As, alen, Bs, blen, s0);
         u64 a_before = s_abefore;
+        u64 b_before = s_bbefore;
         if (UA_WALK2) {
             cnt = tile_walk2<OP>(smem, 0, alen, (int)(Bs - smem), blen, a_before,
-                                 has_ab, has_bn, s0, s1, i0, em, flags);
+                                 has_ab, b_before, has_bb, has_bn, s0, s1, i0, em,
+                                 flags);
         } else {
-            cnt = tile_walk<OP>(As, alen, Bs, blen, a_before, has_ab, has_bn, s0, s1,
-                                i0, em, flags);
+            cnt = tile_walk<OP>(As, alen, Bs, blen, a_before, has_ab, b_before,
+                                has_bb, has_bn, s0, s1, i0, em, flags);
         }
         if (MODE == MODE_DIRECT) {
             /* OP_MERGE_ALL: every path element emits, so the output position
@@ -574,7 +613,7 @@ __device__ __forceinline__ TileMeta d_tile_meta(const UaDesc *__restrict__ descs
 #define UA_PIPE_REGS (UA_WPT)
 __device__ __forceinline__ void d_reg_load(const TileMeta &m, int tid,
                                            u64 (&val)[UA_PIPE_REGS], u64 &abefore,
-                                           u64 &bnext) {
+                                           u64 &bbefore, u64 &bnext) {
 #pragma unroll
     for (int k = 0; k < UA_PIPE_REGS; k++) {
         int e = tid + k * UA_BLOCK;
@@ -585,13 +624,15 @@ __device__ __forceinline__ void d_reg_load(const TileMeta &m, int tid,
     }
     if (tid == 0) {
         abefore = m.has_ab ? m.d.u[m.a0 - 1] : 0;
+        bbefore = (m.b0 > 0) ? m.d.v[m.b0 - 1] : 0;
         bnext = m.has_bn ? m.d.v[m.b0 + m.blen] : 0;
     }
 }
 
 __device__ __forceinline__ void d_reg_commit(const TileMeta &m, int tid, u64 *smem,
                                              const u64 (&val)[UA_PIPE_REGS],
-                                             u64 abefore, u64 bnext, u64 *s_abefore) {
+                                             u64 abefore, u64 bbefore, u64 bnext,
+                                             u64 *s_abefore, u64 *s_bbefore) {
     int boff = (m.alen + 1) & ~1;
 #pragma unroll
     for (int k = 0; k < UA_PIPE_REGS; k++) {
@@ -601,13 +642,14 @@ __device__ __forceinline__ void d_reg_commit(const TileMeta &m, int tid, u64 *sm
     }
     if (tid == 0) {
         *s_abefore = abefore;
+        *s_bbefore = bbefore;
         smem[boff + m.blen] = bnext; /* lookahead slot (garbage if !has_bn) */
     }
 }
 
 template <int OP, int MODE>
 __device__ __forceinline__ void d_tile_body(const TileMeta &m, u64 t, int tid,
-                                            const u64 *smem, u64 a_before,
+                                            const u64 *smem, u64 a_before, u64 b_before,
                                             u64 *__restrict__ staging, u64 stage_stride,
                                             u32 *__restrict__ tile_cnt,
                                             const u64 *__restrict__ offs,
@@ -625,8 +667,8 @@ __device__ __forceinline__ void d_tile_body(const TileMeta &m, u64 t, int tid,
     int cnt = 0;
     if (m.valid) {
         int i0 = d_merge_path_lds(smem, alen, smem + boff, blen, s0);
-        cnt = tile_walk2<OP>(smem, 0, alen, boff, blen, a_before, m.has_ab, m.has_bn,
-                             s0, s1, i0, em, flags);
+        cnt = tile_walk2<OP>(smem, 0, alen, boff, blen, a_before, m.has_ab, b_before,
+                             m.b0 > 0, m.has_bn, s0, s1, i0, em, flags);
     }
     u32 excl, total;
     d_block_scan(tid, (u32)cnt, scanbuf, excl, total);
@@ -661,6 +703,7 @@ __global__ __launch_bounds__(UA_BLOCK) void k_tiles_pipe(
     __shared__ __align__(16) u64 smem[2][UA_TILE + 4];
     __shared__ u32 scanbuf[2][UA_BLOCK / 64];
     __shared__ u64 s_abefore[2];
+    __shared__ u64 s_bbefore[2];
 
     int tid = threadIdx.x;
     u64 t0 = (u64)blockIdx.x * 2;
@@ -668,22 +711,22 @@ __global__ __launch_bounds__(UA_BLOCK) void k_tiles_pipe(
     TileMeta m0 = d_tile_meta(descs, tile_pair, tile_a0, total_tiles, t0);
     TileMeta m1 = d_tile_meta(descs, tile_pair, tile_a0, total_tiles, t1);
 
-    u64 v0[UA_PIPE_REGS], ab0 = 0, bn0 = 0;
-    d_reg_load(m0, tid, v0, ab0, bn0);
-    d_reg_commit(m0, tid, smem[0], v0, ab0, bn0, &s_abefore[0]);
+    u64 v0[UA_PIPE_REGS], ab0 = 0, bb0 = 0, bn0 = 0;
+    d_reg_load(m0, tid, v0, ab0, bb0, bn0);
+    d_reg_commit(m0, tid, smem[0], v0, ab0, bb0, bn0, &s_abefore[0], &s_bbefore[0]);
     /* issue tile-1 loads NOW: they stay in flight across the barrier and
      * the tile-0 walk */
-    u64 v1[UA_PIPE_REGS], ab1 = 0, bn1 = 0;
-    if (m1.valid) d_reg_load(m1, tid, v1, ab1, bn1);
+    u64 v1[UA_PIPE_REGS], ab1 = 0, bb1 = 0, bn1 = 0;
+    if (m1.valid) d_reg_load(m1, tid, v1, ab1, bb1, bn1);
     __syncthreads();
 
-    d_tile_body<OP, MODE>(m0, t0, tid, smem[0], s_abefore[0], staging, stage_stride,
-                          tile_cnt, offs, partials, scanbuf[0]);
+    d_tile_body<OP, MODE>(m0, t0, tid, smem[0], s_abefore[0], s_bbefore[0], staging,
+                          stage_stride, tile_cnt, offs, partials, scanbuf[0]);
     if (!m1.valid) return;
-    d_reg_commit(m1, tid, smem[1], v1, ab1, bn1, &s_abefore[1]);
+    d_reg_commit(m1, tid, smem[1], v1, ab1, bb1, bn1, &s_abefore[1], &s_bbefore[1]);
     __syncthreads();
-    d_tile_body<OP, MODE>(m1, t1, tid, smem[1], s_abefore[1], staging, stage_stride,
-                          tile_cnt, offs, partials, scanbuf[1]);
+    d_tile_body<OP, MODE>(m1, t1, tid, smem[1], s_abefore[1], s_bbefore[1], staging,
+                          stage_stride, tile_cnt, offs, partials, scanbuf[1]);
 }
 
 /* ==================== kernel: bitonic chunk sort (segmented sort stage 1) ====================
@@ -1822,6 +1865,12 @@ extern "C" int ua_merge_k_dev(ua_ctx *c, const uint64_t *const *lists,
     if (k <= 0) {
         *out_n = 0;
         return UA_OK;
+    }
+    if (k == 1) {
+        /* the reference's heap merge dedups even a single list
+         * (TestMergeSorted9: {1,1,1} -> {1}): union with the empty list */
+        ua_dpair pr = {lists[0], lens[0], lists[0], 0, out};
+        return run_batch_locked(c, &pr, 1, out_n, OP_UNION);
     }
     u64 total = 0;
     for (int i = 0; i < k; i++) total += lens[i];

@@ -78,6 +78,12 @@ GOLDEN_MERGE = [
     ([[], []], []),
     ([[5, 6, 7], [3, 4], [1, 2], []], [1, 2, 3, 4, 5, 6, 7]),
     ([], []),
+    # duplicate-input dedup tables (uidlist_test.go:65,144,151): MergeSorted
+    # collapses ALL duplicates, including within one list
+    ([[11, 13, 16, 18, 20], [12, 14, 15, 15, 16, 16, 17, 25], [1, 2]],
+     [1, 2, 11, 12, 13, 14, 15, 16, 17, 18, 20, 25]),
+    ([[1, 1, 1]], [1]),
+    ([[1, 2, 3, 3, 6], [4, 8, 9]], [1, 2, 3, 4, 6, 8, 9]),
 ]
 
 
@@ -144,6 +150,22 @@ def test_merge_pair_vs_oracle(eng, n, m):
     outs, lens = eng.merge_pairs([to_dev(u)], [to_dev(v)])
     got = to_np(outs[0][:lens[0]])
     assert got.tolist() == orc.merge_sorted([u, v]).tolist()
+
+
+@pytest.mark.parametrize("n,m", [(100, 100), (5000, 3000), (200_000, 100_000)])
+def test_merge_with_duplicates_vs_oracle(eng, n, m):
+    """MergeSorted's dedup is reference-pinned ON duplicate inputs
+    (uidlist_test.go:65,144,151) — sorted-with-dups lists, k-way + pairwise."""
+    rng = np.random.default_rng(SEED + n)
+    u = np.sort(rng.integers(0, (n + m) // 2, size=n, dtype=np.uint64))
+    v = np.sort(rng.integers(0, (n + m) // 2, size=m, dtype=np.uint64))
+    w = np.sort(rng.integers(0, (n + m) // 4, size=m // 2, dtype=np.uint64))
+    outs, lens = eng.merge_pairs([to_dev(u)], [to_dev(v)])
+    assert to_np(outs[0][:lens[0]]).tolist() == orc.merge_sorted([u, v]).tolist()
+    got_k = to_np(eng.merge_sorted([to_dev(u), to_dev(v), to_dev(w)]))
+    assert got_k.tolist() == orc.merge_sorted([u, v, w]).tolist()
+    got_1 = to_np(eng.merge_sorted([to_dev(u)]))
+    assert got_1.tolist() == orc.merge_sorted([u]).tolist()
 
 
 def test_cfg2_planted_overlap(eng):
