@@ -480,7 +480,7 @@ __global__ __launch_bounds__(UA_BLOCK) void k_tiles(
     d_fill_lds(Bs, d.v + b0, blen, tid);
     if (tid == 0) {
         s_abefore = has_ab ? d.u[a0 - 1] : 0;
-        s_bbefore = has_bb ? d.v[b0 - 1] : 0;
+        if (OP == OP_UNION) s_bbefore = has_bb ? d.v[b0 - 1] : 0;
         Bs[blen] = has_bn ? d.v[b1] : 0;
     }
 #endif
@@ -513,7 +513,7 @@ __global__ __launch_bounds__(UA_BLOCK) void k_tiles(
         if (s1 > tilelen) s1 = tilelen;
         int i0 = d_merge_path_lds(As, alen, Bs, blen, s0);
         u64 a_before = s_abefore;
-        u64 b_before = s_bbefore;
+        u64 b_before = (OP == OP_UNION) ? s_bbefore : 0;
         if (UA_WALK2) {
             cnt = tile_walk2<OP>(smem, 0, alen, (int)(Bs - smem), blen, a_before,
                                  has_ab, b_before, has_bb, has_bn, s0, s1, i0, em,
