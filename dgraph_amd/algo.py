@@ -53,6 +53,9 @@ class Engine:
             self._ctx = C.c_void_p()
 
     def __del__(self):
+        import sys
+        if sys.is_finalizing():
+            return  # HIP runtime may already be torn down at interpreter exit
         try:
             self.close()
         except Exception:
@@ -363,6 +366,9 @@ class Batch:
             self._h = C.c_void_p()
 
     def __del__(self):
+        import sys
+        if sys.is_finalizing():
+            return
         try:
             self.close()
         except Exception:
