@@ -84,7 +84,7 @@ def main():
     ap.add_argument("--gpus", type=int, default=1)
     ap.add_argument("--steps", type=int, default=100)
     ap.add_argument("--warmup", type=int, default=10)
-    ap.add_argument("--pairs", type=int, default=96,
+    ap.add_argument("--pairs", type=int, default=192,
                     help="1Mx1M list pairs resident per GPU")
     ap.add_argument("--cpu-seconds", type=float, default=5.0,
                     help="CPU-baseline sample duration (0 disables)")
