@@ -296,6 +296,42 @@ def test_index_of_batch_vs_oracle(eng):
     assert got.tolist() == want
 
 
+def test_randomized_small_batch_fuzz(eng):
+    """512 random small/edge-shaped pairs in ONE batch per op, bit-exact vs
+    the oracle — dense coverage of tile-boundary and shape edge cases."""
+    rng = np.random.default_rng(SEED + 1234)
+    us, vs = [], []
+    for i in range(512):
+        shape = i % 8
+        if shape == 0:
+            n, m = 0, int(rng.integers(0, 50))
+        elif shape == 1:
+            n, m = int(rng.integers(0, 50)), 0
+        elif shape == 2:
+            n, m = 1, int(rng.integers(1, 3000))
+        elif shape == 3:  # straddles one tile boundary
+            n, m = 2048, int(rng.integers(1, 64))
+        elif shape == 4:  # exactly multiple tiles
+            n, m = 2048, 2048
+        else:
+            n, m = int(rng.integers(1, 4000)), int(rng.integers(1, 4000))
+        lim = max(3 * (n + m), 16)
+        us.append(synth.gen_sorted_unique(rng, n, lim))
+        vs.append(synth.gen_sorted_unique(rng, m, lim))
+    d_us = [to_dev(x) for x in us]
+    d_vs = [to_dev(x) for x in vs]
+    i_o, i_l = eng.intersect_pairs(d_us, d_vs)
+    m_o, m_l = eng.merge_pairs(d_us, d_vs)
+    d_o, d_l = eng.difference_pairs(d_us, d_vs)
+    for i in range(512):
+        assert to_np(i_o[i][:i_l[i]]).tolist() == \
+            orc.intersect_with(us[i], vs[i]).tolist(), f"intersect {i}"
+        assert to_np(m_o[i][:m_l[i]]).tolist() == \
+            orc.merge_sorted([us[i], vs[i]]).tolist(), f"merge {i}"
+        assert to_np(d_o[i][:d_l[i]]).tolist() == \
+            orc.difference(us[i], vs[i]).tolist(), f"diff {i}"
+
+
 def test_extreme_values(eng):
     """Boundary values incl. 2^63 crossing and UINT64_MAX (u64 compare, not i64)."""
     u = np.array([0, 1, 2**32, 2**63 - 1, 2**63, 2**64 - 2, 2**64 - 1], dtype=np.uint64)
