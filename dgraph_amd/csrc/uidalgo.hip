@@ -777,6 +777,164 @@ __global__ __launch_bounds__(UA_BLOCK) void k_tiles_pipe(
                           stage_stride, tile_cnt, offs, partials, scanbuf[1]);
 }
 
+
+/* ---- 2-buffer glds pipeline (UA_PIPE2): persistent workgroups grid-stride
+ * over tiles; the NEXT tile's LDS-DMA fill is issued before walking the
+ * current one, and drained only at the iteration-end raw barrier — the
+ * guide's canonical glds 2-buffer overlap.  Scalar head/tail/boundary loads
+ * are ordered BEFORE the glds issue (an ordinary load-result use after a
+ * glds makes hipcc drain vmcnt(0) early — guide §5 trap 4b).  LDS 2x16.4KB
+ * -> 4 WGs/CU. ---- */
+
+struct P2Fill {
+    int aoff, boff;
+};
+
+__device__ __forceinline__ P2Fill d_p2_issue(const TileMeta &m, u64 *buf, int tid,
+                                             u64 *s_ab, u64 *s_bb, int for_union) {
+    P2Fill f{0, 0};
+    if (!m.valid) return f;
+    int ashift = (int)((((uintptr_t)(m.d.u + m.a0)) >> 3) & 1);
+    int bshift = (int)((((uintptr_t)(m.d.v + m.b0)) >> 3) & 1);
+    f.aoff = ashift;
+    f.boff = ((ashift + m.alen + 1) & ~1) + bshift;
+    int wv4 = tid >> 6, lane = tid & 63;
+    int ahead = ashift < m.alen ? ashift : m.alen;
+    int abody = (m.alen - ahead) & ~127;
+    int bhead = bshift < m.blen ? bshift : m.blen;
+    int bbody = (m.blen - bhead) & ~127;
+    /* scalar pieces FIRST (ordinary loads complete before any glds issues) */
+    for (int i = ahead + abody + tid; i < m.alen; i += UA_BLOCK)
+        buf[f.aoff + i] = m.d.u[m.a0 + i];
+    if (tid < ahead) buf[f.aoff + tid] = m.d.u[m.a0 + tid];
+    for (int i = bhead + bbody + tid; i < m.blen; i += UA_BLOCK)
+        buf[f.boff + i] = m.d.v[m.b0 + i];
+    if (tid < bhead) buf[f.boff + tid] = m.d.v[m.b0 + tid];
+    if (tid == 0) {
+        *s_ab = m.has_ab ? m.d.u[m.a0 - 1] : 0;
+        if (for_union) *s_bb = (m.b0 > 0) ? m.d.v[m.b0 - 1] : 0;
+        buf[f.boff + m.blen] = m.has_bn ? m.d.v[m.b0 + m.blen] : 0;
+    }
+    /* LDS-DMA body LAST */
+    for (int e = wv4 * 128; e < abody; e += 4 * 128)
+        __builtin_amdgcn_global_load_lds((const u32 *)(m.d.u + m.a0 + ahead + e + lane * 2),
+                                         (u32 *)&buf[f.aoff + ahead + e], 16, 0, 0);
+    for (int e = wv4 * 128; e < bbody; e += 4 * 128)
+        __builtin_amdgcn_global_load_lds((const u32 *)(m.d.v + m.b0 + bhead + e + lane * 2),
+                                         (u32 *)&buf[f.boff + bhead + e], 16, 0, 0);
+    return f;
+}
+
+/* block scan with a RAW barrier (no vmcnt drain: a prefetch glds may be in
+ * flight during the scan) */
+__device__ __forceinline__ void d_block_scan_raw(int tid, u32 cnt, u32 *wsum,
+                                                 u32 &excl, u32 &total) {
+    int lane = tid & 63, wv = tid >> 6;
+    u32 incl = cnt;
+#pragma unroll
+    for (int o = 1; o < 64; o <<= 1) {
+        u32 x = __shfl_up(incl, o);
+        if (lane >= o) incl += x;
+    }
+    if (lane == 63) wsum[wv] = incl;
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+    u32 wbase = 0;
+#pragma unroll
+    for (int w = 0; w < UA_BLOCK / 64; w++) {
+        u32 sv = wsum[w];
+        if (w < wv) wbase += sv;
+    }
+    total = wsum[0] + wsum[1] + wsum[2] + wsum[3];
+    excl = wbase + incl - cnt;
+}
+
+template <int OP, int MODE>
+__global__ __launch_bounds__(UA_BLOCK) void k_tiles_p2(
+    const UaDesc *__restrict__ descs, const u32 *__restrict__ tile_pair,
+    const u32 *__restrict__ tile_a0, u64 total_tiles,
+    u64 *__restrict__ staging, u64 stage_stride, u32 *__restrict__ tile_cnt,
+    const u64 *__restrict__ offs, const u64 *__restrict__ partials) {
+    __shared__ __align__(16) u64 smem[2][UA_TILE + 4];
+    __shared__ u32 wsum[UA_BLOCK / 64];
+    __shared__ u64 s_ab[2], s_bb[2];
+
+    int tid = threadIdx.x;
+    u64 G = gridDim.x;
+    u64 t = blockIdx.x;
+    if (t >= total_tiles) return;
+
+    TileMeta m = d_tile_meta(descs, tile_pair, tile_a0, total_tiles, t);
+    P2Fill f = d_p2_issue(m, smem[0], tid, &s_ab[0], &s_bb[0], OP == OP_UNION);
+    asm volatile("s_waitcnt vmcnt(0) lgkmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+
+    int cur = 0;
+    for (; t < total_tiles; t += G, cur ^= 1) {
+        /* issue the NEXT tile's fill into the other buffer; it stays in
+         * flight across this tile's walk */
+        u64 tn = t + G;
+        TileMeta mn;
+        P2Fill fn{0, 0};
+        mn.valid = false;
+        if (tn < total_tiles) {
+            mn = d_tile_meta(descs, tile_pair, tile_a0, total_tiles, tn);
+            fn = d_p2_issue(mn, smem[cur ^ 1], tid, &s_ab[cur ^ 1], &s_bb[cur ^ 1],
+                            OP == OP_UNION);
+        }
+
+        u64 *buf = smem[cur];
+        int alen = m.alen, blen = m.blen;
+        int tilelen = alen + blen;
+        int s0 = tid * UA_WPT;
+        int s1 = s0 + UA_WPT;
+        if (s0 > tilelen) s0 = tilelen;
+        if (s1 > tilelen) s1 = tilelen;
+        int i0 = d_merge_path_lds(buf + f.aoff, alen, buf + f.boff, blen, s0);
+        u64 em[UA_WPT];
+        u32 flags;
+        int cnt = tile_walk2<OP>(buf, f.aoff, alen, f.boff, blen, s_ab[cur], m.has_ab,
+                                 s_bb[cur], m.b0 > 0, m.has_bn, s0, s1, i0, em, flags);
+
+        if (MODE == MODE_DIRECT) {
+            u64 lt = t - m.d.tile_base;
+            u64 d0 = lt * UA_TILE;
+            u64 *dst = m.d.out + d0 + (u64)s0;
+            int steps = s1 - s0;
+#pragma unroll
+            for (int q = 0; q < UA_WPT; q++)
+                if (q < steps) dst[q] = em[q];
+        } else {
+            u32 excl, total;
+            d_block_scan_raw(tid, (u32)cnt, wsum, excl, total);
+            if (MODE == MODE_COUNT) {
+                if (tid == 0) tile_cnt[t] = total;
+            } else {
+                u64 *dst;
+                if (MODE == MODE_STAGE) {
+                    dst = staging + t * stage_stride + excl;
+                    if (tid == 0) tile_cnt[t] = total;
+                } else {
+                    dst = m.d.out +
+                          (d_off(offs, partials, t) - d_off(offs, partials, m.d.tile_base)) +
+                          excl;
+                }
+                if (cnt > 0) {
+                    int k = 0;
+#pragma unroll
+                    for (int q = 0; q < UA_WPT; q++)
+                        if (flags & (1u << q)) dst[k++] = em[q];
+                }
+            }
+        }
+        /* drain the prefetch + publish this iteration's LDS writes */
+        asm volatile("s_waitcnt vmcnt(0) lgkmcnt(0)" ::: "memory");
+        __builtin_amdgcn_s_barrier();
+        m = mn;
+        f = fn;
+    }
+}
+
 /* ==================== kernel: bitonic chunk sort (segmented sort stage 1) ====================
  * One workgroup sorts one <=2048-element chunk in LDS (u64 ascending,
  * duplicates kept; padded with UINT64_MAX).  Stage 2 is the merge-path
@@ -1255,6 +1413,9 @@ __global__ __launch_bounds__(UA_BLOCK) void k_enc_finalize(
 #ifndef UA_PIPE
 #define UA_PIPE 0 /* 1 = 2-tile software-pipelined tile kernel (A/B toggle) */
 #endif
+#ifndef UA_PIPE2
+#define UA_PIPE2 0 /* 1 = persistent 2-buffer glds pipeline (A/B toggle) */
+#endif
 
 template <int OP, int MODE>
 static void launch_tiles(ua_ctx *c, const UaDesc *descs, const u32 *tpair,
@@ -1412,6 +1573,15 @@ static void launch_tiles(ua_ctx *c, const UaDesc *descs, const u32 *tpair,
         hipLaunchKernelGGL((k_tiles_pipe<OP, MODE>), dim3((u32)((T + 1) / 2)),
                            dim3(UA_BLOCK), 0, c->stream, descs, tpair, ta0, T, stage,
                            stride, tcnt, offs, part);
+        return;
+    }
+#endif
+#if UA_PIPE2
+    {
+        u64 G = T < 1024 ? T : 1024; /* 4 WGs/CU x 256 CUs resident */
+        hipLaunchKernelGGL((k_tiles_p2<OP, MODE>), dim3((u32)G), dim3(UA_BLOCK), 0,
+                           c->stream, descs, tpair, ta0, T, stage, stride, tcnt, offs,
+                           part);
         return;
     }
 #endif
