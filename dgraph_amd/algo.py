@@ -12,6 +12,7 @@ arrays cross the host boundary bit-for-bit via .view(int64).
 No CPU fallback: every op goes through libuidalgo.so and raises without it.
 """
 import ctypes as C
+import sys
 
 import numpy as np
 
@@ -53,10 +54,9 @@ class Engine:
             self._ctx = C.c_void_p()
 
     def __del__(self):
-        import sys
-        if sys.is_finalizing():
-            return  # HIP runtime may already be torn down at interpreter exit
         try:
+            if sys.is_finalizing():
+                return  # HIP runtime may already be torn down at interpreter exit
             self.close()
         except Exception:
             pass
@@ -366,10 +366,9 @@ class Batch:
             self._h = C.c_void_p()
 
     def __del__(self):
-        import sys
-        if sys.is_finalizing():
-            return
         try:
+            if sys.is_finalizing():
+                return
             self.close()
         except Exception:
             pass
