@@ -118,6 +118,12 @@ def lib():
         L.ua_intersect_packed_batch_dev.argtypes = [
             C.c_void_p, C.c_void_p, C.c_void_p, C.c_void_p, C.c_void_p,
             _u64p, C.c_int, C.POINTER(UaPTask), _u64p]
+        L.ua_pbatch_create.argtypes = [
+            C.c_void_p, C.c_void_p, C.c_void_p, C.c_void_p, C.c_void_p,
+            _u64p, C.c_int, C.POINTER(UaPTask), _voidpp]
+        L.ua_pbatch_run.argtypes = [C.c_void_p, C.c_void_p, _u64p]
+        L.ua_pbatch_destroy.argtypes = [C.c_void_p, C.c_void_p]
+        L.ua_pbatch_destroy.restype = None
         L.ua_decode_dev.argtypes = [C.c_void_p, C.POINTER(UaDPack), _u64, C.c_void_p, _u64p]
         L.ua_encode.argtypes = [_u64p, _u64, C.c_uint32, _voidpp]
         L.ua_encode_dev.argtypes = [C.c_void_p, C.c_void_p, _u64, C.c_uint32,

@@ -270,6 +270,34 @@ class Engine:
             C.c_void_p(dpb.deltas.data_ptr()), pbb, n, tasks, lens))
         return outs, [int(lens[i]) for i in range(n)]
 
+    def make_pack_batch(self, dpb, vs, outs=None, afters=None):
+        """Prepared pack fan-out (ua_pbatch): a standing query plan — tasks
+        and boundaries upload once, each run() is launches only.  vs may
+        repeat one shared tensor (the q.UidList shape)."""
+        import torch
+        from dgraph_amd._lib import UaPTask
+        n = len(dpb.pbb) - 1
+        assert len(vs) == n
+        if afters is None:
+            afters = [0] * n
+        if outs is None:
+            outs = [torch.empty(max(min(dpb.pack_totals[i], vs[i].numel()), 1),
+                                dtype=torch.int64, device=vs[i].device)
+                    for i in range(n)]
+        tasks = (UaPTask * n)()
+        for i in range(n):
+            tasks[i].v = vs[i].data_ptr()
+            tasks[i].m = vs[i].numel()
+            tasks[i].out = outs[i].data_ptr()
+            tasks[i].after_uid = afters[i]
+        pbb = (C.c_uint64 * (n + 1))(*[int(x) for x in dpb.pbb])
+        h = C.c_void_p()
+        check(lib().ua_pbatch_create(
+            self._ctx, C.c_void_p(dpb.bases.data_ptr()),
+            C.c_void_p(dpb.num_uids.data_ptr()), C.c_void_p(dpb.delta_offs.data_ptr()),
+            C.c_void_p(dpb.deltas.data_ptr()), pbb, n, tasks, C.byref(h)))
+        return PackBatch(self, h, n, (dpb, vs, outs))
+
     def intersect_packed(self, dpack, after, v, out=None):
         """algo.IntersectCompressedWith (uidlist.go:33): fused decode+intersect."""
         import torch
@@ -363,6 +391,35 @@ class Batch:
     def close(self):
         if self._h:
             lib().ua_batch_destroy(self._eng._ctx, self._h)
+            self._h = C.c_void_p()
+
+    def __del__(self):
+        try:
+            if sys.is_finalizing():
+                return
+            self.close()
+        except Exception:
+            pass
+
+
+class PackBatch:
+    """Prepared pack fan-out (ua_pbatch)."""
+
+    def __init__(self, engine, handle, n_packs, keepalive):
+        self._eng = engine
+        self._h = handle
+        self.n_packs = n_packs
+        self.outs = keepalive[2]
+        self._keep = keepalive
+
+    def run(self):
+        lens = (C.c_uint64 * self.n_packs)()
+        check(lib().ua_pbatch_run(self._eng._ctx, self._h, lens))
+        return [int(lens[i]) for i in range(self.n_packs)]
+
+    def close(self):
+        if self._h:
+            lib().ua_pbatch_destroy(self._eng._ctx, self._h)
             self._h = C.c_void_p()
 
     def __del__(self):

@@ -2294,6 +2294,129 @@ extern "C" int ua_decode_dev(ua_ctx *c, const ua_dpack *pk, uint64_t seek_uid, u
     return run_packed_locked(c, pk, seek_uid, nullptr, 0, out, out_n, 1);
 }
 
+/* ---- prepared pack fan-out (standing query plan) ---- */
+
+struct ua_pbatch {
+    int n_packs = 0;
+    u64 nb = 0;
+    u64 deltas_bytes = 0;
+    u64 sum_m = 0;
+    const u64 *bases = nullptr;
+    const u32 *nums = nullptr;
+    const u64 *doffs = nullptr;
+    const u8 *deltas = nullptr;
+    void *mem = nullptr; /* pbb + tasks + cnt + offs + partials + pout + staging */
+    u64 *d_pbb = nullptr;
+    ua_ptask *d_tasks = nullptr;
+    u32 *d_cnt = nullptr;
+    u64 *d_offs = nullptr;
+    u64 *d_part = nullptr;
+    u64 *d_pout = nullptr;
+    u64 *d_stage = nullptr;
+    u64 nchunks = 0;
+};
+
+extern "C" int ua_pbatch_create(ua_ctx *c, const uint64_t *bases, const uint32_t *num_uids,
+                                const uint64_t *delta_offs, const uint8_t *deltas,
+                                const uint64_t *pack_block_base, int n_packs,
+                                const ua_ptask *tasks, ua_pbatch **out) {
+    std::lock_guard<std::recursive_mutex> g(c->mu);
+    HIP_TRY(hipSetDevice(c->device));
+    if (n_packs <= 0) return UA_ERR_INVALID;
+    ua_pbatch *b = new ua_pbatch();
+    b->n_packs = n_packs;
+    b->nb = pack_block_base[n_packs];
+    b->bases = bases;
+    b->nums = num_uids;
+    b->doffs = delta_offs;
+    b->deltas = deltas;
+    for (int p = 0; p < n_packs; p++) b->sum_m += tasks[p].m;
+    b->nchunks = (b->nb + 1 + UA_SCAN_CHUNK - 1) / UA_SCAN_CHUNK;
+
+    size_t o_pbb = 0;
+    size_t o_tasks = align16(o_pbb + ((size_t)n_packs + 1) * sizeof(u64));
+    size_t o_cnt = align16(o_tasks + (size_t)n_packs * sizeof(ua_ptask));
+    size_t o_offs = align16(o_cnt + (b->nb + 1) * sizeof(u32));
+    size_t o_part = align16(o_offs + (b->nb + 1) * sizeof(u64));
+    size_t o_pout = align16(o_part + (b->nchunks + 1) * sizeof(u64));
+    size_t o_stage = align16(o_pout + (size_t)n_packs * sizeof(u64));
+    size_t total = align16(o_stage + (b->nb ? b->nb : 1) * UA_MAX_BLOCK_UIDS * sizeof(u64));
+    hipError_t e = hipMalloc(&b->mem, total);
+    if (e != hipSuccess) {
+        g_last_hip = e;
+        delete b;
+        return UA_ERR_NOMEM;
+    }
+    u8 *base8 = (u8 *)b->mem;
+    b->d_pbb = (u64 *)(base8 + o_pbb);
+    b->d_tasks = (ua_ptask *)(base8 + o_tasks);
+    b->d_cnt = (u32 *)(base8 + o_cnt);
+    b->d_offs = (u64 *)(base8 + o_offs);
+    b->d_part = (u64 *)(base8 + o_part);
+    b->d_pout = (u64 *)(base8 + o_pout);
+    b->d_stage = (u64 *)(base8 + o_stage);
+
+    std::vector<u8> hostbuf(o_cnt);
+    memcpy(hostbuf.data(), pack_block_base, ((size_t)n_packs + 1) * sizeof(u64));
+    memcpy(hostbuf.data() + o_tasks, tasks, (size_t)n_packs * sizeof(ua_ptask));
+    HIP_TRY(hipMemcpyAsync(b->mem, hostbuf.data(), hostbuf.size(),
+                           hipMemcpyHostToDevice, c->stream));
+    HIP_TRY(hipMemsetAsync(b->d_cnt + b->nb, 0, sizeof(u32), c->stream));
+    u64 db = 0;
+    HIP_TRY(hipMemcpyAsync(&db, delta_offs + b->nb, sizeof(u64),
+                           hipMemcpyDeviceToHost, c->stream));
+    HIP_TRY(hipStreamSynchronize(c->stream));
+    b->deltas_bytes = db;
+    *out = b;
+    return UA_OK;
+}
+
+extern "C" void ua_pbatch_destroy(ua_ctx *c, ua_pbatch *b) {
+    if (!b) return;
+    (void)hipSetDevice(c->device);
+    if (b->mem) (void)hipFree(b->mem);
+    delete b;
+}
+
+extern "C" int ua_pbatch_run(ua_ctx *c, ua_pbatch *b, uint64_t *out_lens) {
+    std::lock_guard<std::recursive_mutex> g(c->mu);
+    HIP_TRY(hipSetDevice(c->device));
+    u64 nb = b->nb;
+    if (nb > 0) {
+        u64 nwg = (nb + UA_PKW - 1) / UA_PKW;
+        HIP_TRY(hipEventRecord(c->ev[0], c->stream));
+        hipLaunchKernelGGL(k_packed<0>, dim3((u32)nwg), dim3(UA_BLOCK), 0, c->stream,
+                           b->bases, b->nums, b->doffs, b->deltas, nb, (u64)0,
+                           (const u64 *)nullptr, (u64)0, b->d_stage, b->d_cnt,
+                           b->d_pbb, b->n_packs, b->d_tasks);
+        HIP_TRY(hipEventRecord(c->ev[1], c->stream));
+        hipLaunchKernelGGL(k_scan1, dim3((u32)b->nchunks), dim3(UA_BLOCK), 0, c->stream,
+                           b->d_cnt, nb + 1, b->d_offs, b->d_part);
+        hipLaunchKernelGGL(k_scan2, dim3(1), dim3(UA_BLOCK), 0, c->stream, b->d_part,
+                           b->nchunks);
+        hipLaunchKernelGGL(k_compact_pack, dim3((u32)((nb + 15) / 16)), dim3(UA_BLOCK),
+                           0, c->stream, b->d_tasks, b->d_pbb, b->n_packs, b->d_cnt,
+                           b->d_offs, b->d_part, b->d_stage, nb);
+    }
+    u64 poutblk = ((u64)b->n_packs + UA_BLOCK - 1) / UA_BLOCK;
+    hipLaunchKernelGGL(k_pair_out, dim3((u32)poutblk), dim3(UA_BLOCK), 0, c->stream,
+                       b->d_offs, b->d_part, b->d_pbb, b->n_packs, b->d_pout);
+    HIP_TRY(hipMemcpyAsync(out_lens, b->d_pout, (size_t)b->n_packs * sizeof(u64),
+                           hipMemcpyDeviceToHost, c->stream));
+    HIP_TRY(hipStreamSynchronize(c->stream));
+    HIP_TRY(hipGetLastError());
+    if (nb > 0) {
+        float ms = 0.f;
+        HIP_TRY(hipEventElapsedTime(&ms, c->ev[0], c->ev[1]));
+        c->kernel_ms += ms;
+        c->n_launches += 1;
+    }
+    u64 out_elems = 0;
+    for (int p = 0; p < b->n_packs; p++) out_elems += out_lens[p];
+    c->bytes_algo += b->deltas_bytes + nb * 20 + 8 * (b->sum_m + out_elems);
+    return UA_OK;
+}
+
 /* ---- GPU codec.Encode pipeline (codec.go:393 semantics, engine layout) ---- */
 extern "C" int ua_encode_dev(ua_ctx *c, const uint64_t *uids, uint64_t n,
                              uint32_t block_size, uint64_t *bases, uint32_t *num_uids,

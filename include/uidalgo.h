@@ -211,6 +211,20 @@ int ua_intersect_packed_batch_dev(ua_ctx *, const uint64_t *bases,
                                   int n_packs, const ua_ptask *tasks /* host */,
                                   uint64_t *out_lens);
 
+/* Prepared pack fan-out (a standing query plan): task/boundary upload and
+ * workspace sizing once at create; each run is launches + one lens read.
+ * The flat pack arena and the tasks' v/out buffers are caller-owned device
+ * memory that must outlive the handle; their contents may change between
+ * runs (the plan caches no data-dependent state, unlike ua_batch). */
+typedef struct ua_pbatch ua_pbatch;
+int ua_pbatch_create(ua_ctx *, const uint64_t *bases, const uint32_t *num_uids,
+                     const uint64_t *delta_offs, const uint8_t *deltas,
+                     const uint64_t *pack_block_base /* host [n_packs+1] */,
+                     int n_packs, const ua_ptask *tasks /* host */,
+                     ua_pbatch **out);
+int ua_pbatch_run(ua_ctx *, ua_pbatch *, uint64_t *out_lens);
+void ua_pbatch_destroy(ua_ctx *, ua_pbatch *);
+
 /* ---- host-pointer convenience (upload+compute+download; mirrors the algo
  * package signatures 1:1 for the cgo shim) ---- */
 int ua_intersect(ua_ctx *, const uint64_t *u, uint64_t n, const uint64_t *v,

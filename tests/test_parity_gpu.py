@@ -496,6 +496,37 @@ def test_packed_fanout_batch_vs_oracle(eng):
         assert got.tolist() == want.tolist(), f"pack {i}"
 
 
+def test_prepared_pack_batch_vs_oneshot(eng):
+    """ua_pbatch (standing query plan): repeated runs equal the one-shot
+    batched call; v contents may change between runs."""
+    import torch
+    rng = np.random.default_rng(SEED + 321)
+    packs_np, flat = [], []
+    for _ in range(32):
+        uids = np.unique(synth.getuids_geometric(rng, int(rng.integers(1, 40_000))))
+        packs_np.append(uids)
+        flat.append(algo.encode_flat(uids, 256))
+    shared = synth.gen_sorted_unique(rng, 30_000, 1_500_000)
+    d_shared = to_dev(shared)
+    dpb = eng.upload_pack_batch(flat)
+
+    pb = eng.make_pack_batch(dpb, [d_shared] * 32)
+    ref_outs, ref_lens = eng.intersect_packed_batch(dpb, [d_shared] * 32)
+    for _ in range(3):
+        lens = pb.run()
+        for i in range(32):
+            assert lens[i] == ref_lens[i]
+            assert torch.equal(pb.outs[i][:lens[i]], ref_outs[i][:ref_lens[i]])
+    # mutate the shared v in place: the plan has no data-dependent cache
+    shared2 = synth.gen_sorted_unique(rng, 30_000, 1_500_000)
+    d_shared.copy_(torch.from_numpy(shared2.view(np.int64)).cuda())
+    lens2 = pb.run()
+    for i in [0, 15, 31]:
+        want = orc.intersect_compressed_with(orc.Pack(packs_np[i], 256), 0, shared2)
+        assert to_np(pb.outs[i][:lens2[i]]).tolist() == want.tolist()
+    pb.close()
+
+
 def test_host_packed_api(eng):
     """ua_intersect_packed host-pointer convenience (the cgo surface)."""
     import ctypes as Ct
