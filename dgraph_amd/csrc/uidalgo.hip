@@ -462,6 +462,7 @@ __global__ __launch_bounds__(UA_BLOCK) void k_tiles(
     u64 lt = t - d.tile_base;
     u64 path = d.n + d.m;
     u64 d0 = lt * UA_TILE;
+    if (d0 > path) d0 = path; /* capacity-tiled batches can overshoot the path */
     u64 d1 = d0 + UA_TILE;
     if (d1 > path) d1 = path;
     u32 a0 = tile_a0[t];
@@ -643,6 +644,7 @@ __device__ __forceinline__ TileMeta d_tile_meta(const UaDesc *__restrict__ descs
     u64 lt = t - m.d.tile_base;
     u64 path = m.d.n + m.d.m;
     u64 d0 = lt * UA_TILE;
+    if (d0 > path) d0 = path;
     u64 d1 = d0 + UA_TILE;
     if (d1 > path) d1 = path;
     u32 a1 = (t + 1 < total_tiles && tile_pair[t + 1] == m.p) ? tile_a0[t + 1]
@@ -1410,6 +1412,27 @@ __global__ __launch_bounds__(UA_BLOCK) void k_enc_finalize(
 }
 
 
+/* ---- device-chained merge tree helpers ---- */
+
+/* fill a round's descriptor lengths from the previous round's device lens */
+__global__ __launch_bounds__(UA_BLOCK) void k_make_descs(
+    UaDesc *__restrict__ descs, const u64 *__restrict__ lens_prev, int nk_prev,
+    int npair) {
+    int j = blockIdx.x * UA_BLOCK + threadIdx.x;
+    if (j >= npair) return;
+    descs[j].n = lens_prev[2 * j];
+    descs[j].m = (2 * j + 1 < nk_prev) ? lens_prev[2 * j + 1] : 0;
+}
+
+/* copy src[0..*len_ptr) to out without a host round-trip for the length */
+__global__ __launch_bounds__(UA_BLOCK) void k_copy_len(
+    u64 *__restrict__ out, const u64 *__restrict__ src, const u64 *__restrict__ len_ptr) {
+    u64 n = *len_ptr;
+    for (u64 i = (u64)blockIdx.x * UA_BLOCK + threadIdx.x; i < n;
+         i += (u64)gridDim.x * UA_BLOCK)
+        out[i] = src[i];
+}
+
 #ifndef UA_PIPE
 #define UA_PIPE 0 /* 1 = 2-tile software-pipelined tile kernel (A/B toggle) */
 #endif
@@ -2090,55 +2113,156 @@ extern "C" int ua_merge_k_dev(ua_ctx *c, const uint64_t *const *lists,
         ua_dpair pr = {lists[0], lens[0], lists[0], 0, out};
         return run_batch_locked(c, &pr, 1, out_n, OP_UNION);
     }
+    /* Device-chained pairwise union tree: positions and tile layouts are
+     * CAPACITY-based (host-known up front), while each round's actual pair
+     * lengths are filled on-device from the previous round's outputs
+     * (k_make_descs) — the whole log2(k)-round tree enqueues with a single
+     * final sync.  Tail tiles past a pair's actual path no-op in k_tiles.
+     * Odd items pair with an empty side (a dedup copy), so every item moves
+     * to the other ping-pong buffer each round. */
+    HIP_TRY(hipSetDevice(c->device));
     u64 total = 0;
     for (int i = 0; i < k; i++) total += lens[i];
     int rc;
-    {
-        std::lock_guard<std::recursive_mutex> g(c->mu);
-        if ((rc = ws_reserve(c, WS_SCRATCH_A, (total ? total : 1) * sizeof(u64)))) return rc;
-        if ((rc = ws_reserve(c, WS_SCRATCH_B, (total ? total : 1) * sizeof(u64)))) return rc;
-    }
-    std::vector<const u64 *> cur_ptr(lists, lists + k);
-    std::vector<u64> cur_len(lens, lens + k);
+    if ((rc = ws_reserve(c, WS_SCRATCH_A, (total ? total : 1) * sizeof(u64)))) return rc;
+    if ((rc = ws_reserve(c, WS_SCRATCH_B, (total ? total : 1) * sizeof(u64)))) return rc;
     u64 *bufs[2] = {(u64 *)c->ws[WS_SCRATCH_A], (u64 *)c->ws[WS_SCRATCH_B]};
+
+    /* host planning over capacities */
+    struct Round {
+        int nk_prev, npair;
+        u64 total_tiles;
+        size_t desc_off, tb_off; /* element offsets into the concat uploads */
+    };
+    std::vector<Round> rounds;
+    std::vector<UaDesc> all_descs;
+    std::vector<u64> all_tb;
+    std::vector<u64> caps(lens, lens + k);
+    std::vector<const u64 *> ptrs(lists, lists + k);
+    u64 max_tiles = 0, cap_work = 0;
     int which = 0;
-    while (cur_ptr.size() > 1) {
-        int nk = (int)cur_ptr.size();
-        int npair = nk / 2;
-        std::vector<ua_dpair> prs(npair);
-        std::vector<u64> olens(npair);
-        u64 off = 0;
+    while ((int)caps.size() > 1) {
+        int nk = (int)caps.size();
+        int npair = (nk + 1) / 2;
+        Round r;
+        r.nk_prev = nk;
+        r.npair = npair;
+        r.desc_off = all_descs.size();
+        r.tb_off = all_tb.size();
         u64 *buf = bufs[which];
-        for (int p = 0; p < npair; p++) {
-            prs[p] = {cur_ptr[2 * p], cur_len[2 * p], cur_ptr[2 * p + 1], cur_len[2 * p + 1],
-                      buf + off};
-            off += cur_len[2 * p] + cur_len[2 * p + 1];
+        std::vector<u64> ncaps;
+        std::vector<const u64 *> nptrs;
+        u64 off = 0, tiles = 0;
+        for (int pj = 0; pj < npair; pj++) {
+            u64 ca = caps[2 * pj];
+            u64 cb = (2 * pj + 1 < nk) ? caps[2 * pj + 1] : 0;
+            const u64 *pa = ptrs[2 * pj];
+            const u64 *pb = (2 * pj + 1 < nk) ? ptrs[2 * pj + 1] : ptrs[2 * pj];
+            all_tb.push_back(tiles);
+            /* n,m are placeholders; k_make_descs overwrites from device lens */
+            all_descs.push_back({pa, ca, pb, cb, buf + off, tiles});
+            tiles += (ca + cb + UA_TILE - 1) / UA_TILE;
+            nptrs.push_back(buf + off);
+            ncaps.push_back(ca + cb);
+            off += ca + cb;
+            cap_work += ca + cb;
         }
-        if ((rc = run_batch(c, prs.data(), npair, olens.data(), OP_UNION))) return rc;
-        std::vector<const u64 *> nxt_ptr;
-        std::vector<u64> nxt_len;
-        off = 0;
-        for (int p = 0; p < npair; p++) {
-            nxt_ptr.push_back(buf + off);
-            nxt_len.push_back(olens[p]);
-            off += cur_len[2 * p] + cur_len[2 * p + 1];
-        }
-        if (nk % 2) { /* odd list carries over */
-            nxt_ptr.push_back(cur_ptr[nk - 1]);
-            nxt_len.push_back(cur_len[nk - 1]);
-        }
-        cur_ptr.swap(nxt_ptr);
-        cur_len.swap(nxt_len);
+        all_tb.push_back(tiles);
+        r.total_tiles = tiles;
+        if (tiles > max_tiles) max_tiles = tiles;
+        rounds.push_back(r);
+        caps.swap(ncaps);
+        ptrs.swap(nptrs);
         which ^= 1;
     }
+    int R = (int)rounds.size();
+
+    /* workspace: concat descs+tb+lens ping-pong in WS_DESC; tiles in the
+     * usual slots sized by the largest round */
+    size_t descs_bytes = all_descs.size() * sizeof(UaDesc);
+    size_t tb_bytes = all_tb.size() * sizeof(u64);
+    size_t lens_bytes = (size_t)k * sizeof(u64);
+    if ((rc = ws_reserve(c, WS_DESC, descs_bytes + tb_bytes + 2 * lens_bytes + 64)))
+        return rc;
+    u8 *wbase = (u8 *)c->ws[WS_DESC];
+    UaDesc *d_descs_all = (UaDesc *)wbase;
+    u64 *d_tb_all = (u64 *)(wbase + descs_bytes);
+    u64 *d_lens[2] = {(u64 *)(wbase + descs_bytes + tb_bytes),
+                      (u64 *)(wbase + descs_bytes + tb_bytes + lens_bytes)};
     {
-        std::lock_guard<std::recursive_mutex> g(c->mu);
-        HIP_TRY(hipSetDevice(c->device));
-        HIP_TRY(hipMemcpyAsync(out, cur_ptr[0], cur_len[0] * sizeof(u64),
-                               hipMemcpyDeviceToDevice, c->stream));
-        HIP_TRY(hipStreamSynchronize(c->stream));
+        std::vector<u8> hostbuf(descs_bytes + tb_bytes + lens_bytes);
+        memcpy(hostbuf.data(), all_descs.data(), descs_bytes);
+        memcpy(hostbuf.data() + descs_bytes, all_tb.data(), tb_bytes);
+        memcpy(hostbuf.data() + descs_bytes + tb_bytes, lens, lens_bytes);
+        HIP_TRY(hipMemcpyAsync(wbase, hostbuf.data(), hostbuf.size(),
+                               hipMemcpyHostToDevice, c->stream));
+        HIP_TRY(hipStreamSynchronize(c->stream)); /* hostbuf is scoped */
     }
-    *out_n = cur_len[0];
+    if ((rc = ws_reserve(c, WS_TPAIR, (max_tiles + 1) * sizeof(u32)))) return rc;
+    if ((rc = ws_reserve(c, WS_TA0, (max_tiles + 1) * sizeof(u32)))) return rc;
+    if ((rc = ws_reserve(c, WS_TCNT, (max_tiles + 1) * sizeof(u32)))) return rc;
+    if ((rc = ws_reserve(c, WS_TOFF, (max_tiles + 1) * sizeof(u64)))) return rc;
+    u64 max_chunks = (max_tiles + 1 + UA_SCAN_CHUNK - 1) / UA_SCAN_CHUNK;
+    if ((rc = ws_reserve(c, WS_PARTIAL, (max_chunks + 1) * sizeof(u64)))) return rc;
+    u32 *d_tpair = (u32 *)c->ws[WS_TPAIR];
+    u32 *d_ta0 = (u32 *)c->ws[WS_TA0];
+    u32 *d_tcnt = (u32 *)c->ws[WS_TCNT];
+    u64 *d_toff = (u64 *)c->ws[WS_TOFF];
+    u64 *d_part = (u64 *)c->ws[WS_PARTIAL];
+
+    HIP_TRY(hipEventRecord(c->ev[0], c->stream));
+    int lcur = 0;
+    for (int r = 0; r < R; r++) {
+        const Round &rd = rounds[r];
+        UaDesc *d_descs = d_descs_all + rd.desc_off;
+        u64 *d_tb = d_tb_all + rd.tb_off;
+        u64 T = rd.total_tiles;
+        u64 mk = ((u64)rd.npair + UA_BLOCK - 1) / UA_BLOCK;
+        hipLaunchKernelGGL(k_make_descs, dim3((u32)mk), dim3(UA_BLOCK), 0, c->stream,
+                           d_descs, d_lens[lcur], rd.nk_prev, rd.npair);
+        HIP_TRY(hipMemsetAsync(d_tcnt + T, 0, sizeof(u32), c->stream));
+        if (T > 0) {
+            u64 pblk = (T + UA_BLOCK - 1) / UA_BLOCK;
+            hipLaunchKernelGGL(k_partition, dim3((u32)pblk), dim3(UA_BLOCK), 0, c->stream,
+                               d_descs, d_tb, rd.npair, T, d_tpair, d_ta0, 0);
+            hipLaunchKernelGGL(k_partition, dim3((u32)pblk), dim3(UA_BLOCK), 0, c->stream,
+                               d_descs, d_tb, rd.npair, T, d_tpair, d_ta0, 1);
+            launch_tiles<OP_UNION, MODE_COUNT>(c, d_descs, d_tpair, d_ta0, T, nullptr,
+                                               0, d_tcnt, nullptr, nullptr);
+            u64 nchunks = (T + 1 + UA_SCAN_CHUNK - 1) / UA_SCAN_CHUNK;
+            hipLaunchKernelGGL(k_scan1, dim3((u32)nchunks), dim3(UA_BLOCK), 0, c->stream,
+                               d_tcnt, T + 1, d_toff, d_part);
+            hipLaunchKernelGGL(k_scan2, dim3(1), dim3(UA_BLOCK), 0, c->stream, d_part,
+                               nchunks);
+            launch_tiles<OP_UNION, MODE_WRITE>(c, d_descs, d_tpair, d_ta0, T, nullptr,
+                                               0, d_tcnt, d_toff, d_part);
+        } else {
+            HIP_TRY(hipMemsetAsync(d_toff, 0, sizeof(u64), c->stream));
+            HIP_TRY(hipMemsetAsync(d_part, 0, sizeof(u64), c->stream));
+        }
+        u64 poutblk = ((u64)rd.npair + UA_BLOCK - 1) / UA_BLOCK;
+        hipLaunchKernelGGL(k_pair_out, dim3((u32)poutblk), dim3(UA_BLOCK), 0, c->stream,
+                           d_toff, d_part, d_tb, rd.npair, d_lens[lcur ^ 1]);
+        lcur ^= 1;
+    }
+    /* final item: data in ptrs[0] (capacity position), length in d_lens[lcur][0] */
+    u64 copy_grid = ((total ? total : 1) + UA_BLOCK - 1) / UA_BLOCK;
+    if (copy_grid > 2048) copy_grid = 2048;
+    hipLaunchKernelGGL(k_copy_len, dim3((u32)copy_grid), dim3(UA_BLOCK), 0, c->stream,
+                       out, ptrs[0], d_lens[lcur]);
+    u64 final_len = 0;
+    HIP_TRY(hipMemcpyAsync(&final_len, d_lens[lcur], sizeof(u64), hipMemcpyDeviceToHost,
+                           c->stream));
+    HIP_TRY(hipStreamSynchronize(c->stream));
+    HIP_TRY(hipGetLastError());
+    float ms = 0.f;
+    HIP_TRY(hipEventRecord(c->ev[1], c->stream));
+    HIP_TRY(hipStreamSynchronize(c->stream));
+    HIP_TRY(hipEventElapsedTime(&ms, c->ev[0], c->ev[1]));
+    c->kernel_ms += ms;
+    c->n_launches += 2 * R;
+    c->bytes_algo += 8 * 3 * cap_work; /* capacity upper bound (union count+write) */
+    *out_n = final_len;
     return UA_OK;
 }
 
