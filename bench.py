@@ -5,8 +5,9 @@
 One "step" = one pass of the batched-intersect hot path over the resident
 batch: P pairs of 1M x 1M sorted duplicate-free uint64 lists with 1% planted
 overlap (BASELINE.md cfg 2, quoted at N=1), all pairs in ONE grid via the
-C-ABI engine (ua_intersect_batch_dev).  Inputs are resident in HBM before the
-timed region starts.
+C-ABI engine's prepared batch (ua_batch_create/ua_batch_run — descriptors
+and the merge-path partition cached once, steps are launches only).  Inputs
+are resident in HBM before the timed region starts.
 
 N>1 (torchrun, one rank per GPU over RCCL): weak scaling — each rank runs its
 own P pairs; pairs partition embarrassingly, no data-path
