@@ -287,6 +287,26 @@ def test_sort_segments(eng, sizes):
         assert np.array_equal(to_np(t), np.sort(a))
 
 
+def test_trigram_term_combinators(eng):
+    """worker/trigram.go:56-96 + needsIntersect allof/anyof (task.go:303):
+    the index-token combinators are exactly IntersectSorted (allof) and
+    MergeSorted (anyof) over token posting lists — §8f row 4, no new
+    kernels.  8 token lists of skewed sizes, bit-exact vs oracle."""
+    rng = np.random.default_rng(SEED + 888)
+    token_lists = [synth.gen_sorted_unique(rng, int(sz), 500_000)
+                   for sz in [120_000, 90_000, 60_000, 30_000, 8_000, 2_000, 500, 50]]
+    d_lists = [to_dev(t) for t in token_lists]
+    allof = to_np(eng.intersect_sorted(d_lists))       # "allof" fold
+    anyof = to_np(eng.merge_sorted(d_lists))           # "anyof" union
+    assert allof.tolist() == orc.intersect_sorted(token_lists).tolist()
+    assert anyof.tolist() == orc.merge_sorted(token_lists).tolist()
+    # uids ∩ filtered-list shape (trigram.go:87: result ∩ original uid list)
+    uids = synth.gen_sorted_unique(rng, 40_000, 500_000)
+    outs, lens = eng.intersect_pairs([to_dev(uids)], [to_dev(anyof)])
+    assert to_np(outs[0][:lens[0]]).tolist() == \
+        orc.intersect_with(uids, anyof).tolist()
+
+
 def test_index_of_batch_vs_oracle(eng):
     rng = np.random.default_rng(SEED)
     u = synth.gen_sorted_unique(rng, 100_000, 1_000_000)
