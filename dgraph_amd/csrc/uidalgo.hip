@@ -422,6 +422,97 @@ __device__ __forceinline__ int tile_walk2(const u64 *smembase, int aoff, int ale
     return cnt;
 }
 
+/* Dual-chain walk (UA_WALK2X): each thread walks TWO independent
+ * half-segments (diagonals s0 and s0+W/2) with interleaved steps — two
+ * independent LDS-gather dependency chains per thread to close the
+ * latency/MLP gap the SQ counters show (waves park 60% while HBM sits at
+ * ~51%). */
+template <int OP>
+__device__ __forceinline__ int tile_walk2x(const u64 *smembase, int aoff, int alen,
+                                           int boff, int blen, u64 a_before, bool has_ab,
+                                           u64 b_before, bool has_bb, bool has_bn,
+                                           int s0, int s1, int i0a, int i0b, int smid,
+                                           u64 (&em)[UA_WPT], u32 &flags) {
+    const u64 *As = smembase + aoff;
+    const u64 *Bs = smembase + boff;
+    int blen_ext = blen + (has_bn ? 1 : 0);
+    int amax = alen > 0 ? alen - 1 : 0;
+    int bmax = blen_ext > 0 ? blen_ext - 1 : 0;
+    constexpr int H = UA_WPT / 2;
+
+    int ia = i0a, ja = s0 - i0a;
+    int ib = i0b, jb = smid - i0b;
+    int stepsA = smid - s0, stepsB = s1 - smid;
+    u64 aA = As[ia < alen ? ia : amax], bA = Bs[ja < blen_ext ? ja : bmax];
+    u64 aB = As[ib < alen ? ib : amax], bB = Bs[jb < blen_ext ? jb : bmax];
+    bool hpA = false, hpB = false;
+    u64 poA = 0, poB = 0;
+    if (OP == OP_UNION) {
+        poA = d_prev_stream(As, ia, a_before, has_ab, Bs, ja, b_before, has_bb, hpA);
+        poB = d_prev_stream(As, ib, a_before, has_ab, Bs, jb, b_before, has_bb, hpB);
+    }
+    int cnt = 0;
+    flags = 0;
+#pragma unroll
+    for (int q = 0; q < H; q++) {
+        /* chain A step */
+        if (q < stepsA && (ia < alen || ja < blen)) {
+            bool inA = ia < alen, inB = ja < blen;
+            bool takeA = inA && (!inB || aA <= bA);
+            bool eq = (aA == bA) && (ja < blen_ext);
+            bool emit;
+            u64 val = takeA ? aA : bA;
+            if (OP == OP_INTERSECT) emit = takeA && eq;
+            else if (OP == OP_DIFF) emit = takeA && !eq;
+            else if (OP == OP_MERGE_ALL) emit = true;
+            else {
+                emit = !hpA || val != poA;
+                poA = val;
+                hpA = true;
+            }
+            em[q] = (OP == OP_INTERSECT || OP == OP_DIFF) ? aA : val;
+            flags |= ((u32)emit) << q;
+            cnt += emit;
+            int ni = ia + (takeA ? 1 : 0), nj = ja + (takeA ? 0 : 1);
+            int raddr = takeA ? (aoff + (ni < alen ? ni : amax))
+                              : (boff + (nj < blen_ext ? nj : bmax));
+            u64 r = smembase[raddr];
+            aA = takeA ? r : aA;
+            bA = takeA ? bA : r;
+            ia = ni;
+            ja = nj;
+        }
+        /* chain B step (independent registers: interleaves with A's gather) */
+        if (q < stepsB && (ib < alen || jb < blen)) {
+            bool inA = ib < alen, inB = jb < blen;
+            bool takeA = inA && (!inB || aB <= bB);
+            bool eq = (aB == bB) && (jb < blen_ext);
+            bool emit;
+            u64 val = takeA ? aB : bB;
+            if (OP == OP_INTERSECT) emit = takeA && eq;
+            else if (OP == OP_DIFF) emit = takeA && !eq;
+            else if (OP == OP_MERGE_ALL) emit = true;
+            else {
+                emit = !hpB || val != poB;
+                poB = val;
+                hpB = true;
+            }
+            em[H + q] = (OP == OP_INTERSECT || OP == OP_DIFF) ? aB : val;
+            flags |= ((u32)emit) << (H + q);
+            cnt += emit;
+            int ni = ib + (takeA ? 1 : 0), nj = jb + (takeA ? 0 : 1);
+            int raddr = takeA ? (aoff + (ni < alen ? ni : amax))
+                              : (boff + (nj < blen_ext ? nj : bmax));
+            u64 r = smembase[raddr];
+            aB = takeA ? r : aB;
+            bB = takeA ? bB : r;
+            ib = ni;
+            jb = nj;
+        }
+    }
+    return cnt;
+}
+
 /* cooperative global->LDS fill, 16-B vectorized on the aligned body
  * (8-B/lane loads cap ~60% of the dwordx4 HBM rate — guide §2/G13) */
 __device__ __forceinline__ void d_fill_lds(u64 *dst, const u64 *__restrict__ src,
@@ -563,7 +654,19 @@ __global__ __launch_bounds__(UA_BLOCK) void k_tiles(
         int i0 = d_merge_path_lds(As, alen, Bs, blen, s0);
         u64 a_before = s_abefore;
         u64 b_before = (OP == OP_UNION) ? s_bbefore : 0;
-        if (UA_WALK2) {
+#ifndef UA_WALK2X
+#define UA_WALK2X 0 /* 1 = dual-chain walk (2 independent gather chains/thread) */
+#endif
+        if (UA_WALK2X) {
+            int smid = s0 + UA_WPT / 2;
+            if (smid > tilelen) smid = tilelen;
+            if (smid < s0) smid = s0;
+            if (smid > s1) smid = s1;
+            int i0b = d_merge_path_lds(As, alen, Bs, blen, smid);
+            cnt = tile_walk2x<OP>(smem, (int)(As - smem), alen, (int)(Bs - smem), blen,
+                                  a_before, has_ab, b_before, has_bb, has_bn, s0, s1,
+                                  i0, i0b, smid, em, flags);
+        } else if (UA_WALK2) {
             cnt = tile_walk2<OP>(smem, (int)(As - smem), alen, (int)(Bs - smem), blen,
                                  a_before, has_ab, b_before, has_bb, has_bn, s0, s1,
                                  i0, em, flags);
