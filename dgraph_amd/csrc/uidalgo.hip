@@ -181,7 +181,66 @@ __global__ __launch_bounds__(UA_BLOCK) void k_partition(
 /* ==================== kernel: tile set-algebra ==================== */
 
 enum { OP_INTERSECT = 0, OP_UNION = 1, OP_DIFF = 2, OP_MERGE_ALL = 3 };
-enum { MODE_STAGE = 0, MODE_COUNT = 1, MODE_WRITE = 2, MODE_DIRECT = 3 };
+enum { MODE_STAGE = 0, MODE_COUNT = 1, MODE_WRITE = 2, MODE_DIRECT = 3,
+       MODE_LOOKBACK = 4 };
+
+/* ---- decoupled-lookback tile prefix (MODE_LOOKBACK) ----
+ * Single-pass pipeline: each tile walks once, publishes its emission count,
+ * resolves its pair-local exclusive prefix by looking back over predecessor
+ * tiles (CUB-style: 64 lanes inspect 64 predecessors per round), and writes
+ * its emissions at final positions — no staging buffer, no separate scan /
+ * compact / pair_out launches, and union needs ONE walk instead of
+ * count+write.  Safe on CDNA4: workgroups dispatch first->last
+ * (MI355X_MICROARCH.md §dispatch; the same assumption rocPRIM's device scan
+ * ships on), and the 8-byte flag word is an untorn agent-scope (sc1)
+ * load/store granule, so no fences are needed — the word itself carries the
+ * value.
+ *
+ * Flag word: [63:48] generation (stale-run guard, wraps via memset),
+ * [47:46] status (0 invalid / 1 aggregate / 2 pair-local inclusive prefix),
+ * [45:0] value. */
+#define UA_LB_GEN_MAX 0xFFFFull
+__device__ __forceinline__ u64 d_lb_word(u64 gen, u64 status, u64 val) {
+    return (gen << 48) | (status << 46) | val;
+}
+
+/* wave0 (tid<64) resolves the pair-local exclusive prefix of tile t */
+__device__ __forceinline__ u64 d_lb_resolve(u64 *lbf, u64 gen, u64 t, u64 base_t,
+                                            int lane) {
+    u64 run = 0;
+    u64 hi = t;
+    for (;;) {
+        long idx = (long)hi - 1 - lane;
+        int st = 2;
+        u64 val = 0;
+        if (idx >= (long)base_t) {
+            u64 w = __hip_atomic_load(&lbf[idx], __ATOMIC_RELAXED,
+                                      __HIP_MEMORY_SCOPE_AGENT);
+            while ((w >> 48) != gen || !((w >> 46) & 3)) {
+                __builtin_amdgcn_s_sleep(1);
+                w = __hip_atomic_load(&lbf[idx], __ATOMIC_RELAXED,
+                                      __HIP_MEMORY_SCOPE_AGENT);
+            }
+            st = (int)((w >> 46) & 3);
+            val = w & ((1ull << 46) - 1);
+        }
+        u64 pm = __ballot(st == 2);
+        /* nearest PREFIX lane (smallest back distance); lanes past it add
+         * nothing.  pm==0: whole window is aggregates — sum and slide. */
+        u64 contrib;
+        if (pm) {
+            int k = __ffsll((unsigned long long)pm) - 1;
+            contrib = (lane <= k) ? val : 0;
+        } else {
+            contrib = val;
+        }
+#pragma unroll
+        for (int o = 32; o >= 1; o >>= 1) contrib += __shfl_xor(contrib, o);
+        run += contrib;
+        if (pm) return run;
+        hi -= 64;
+    }
+}
 
 /* One pass over the thread's merge-path segment.  Emissions are recorded in
  * a statically-indexed register array (em[s], flag bit s) so the walk runs
@@ -288,12 +347,14 @@ __device__ __forceinline__ void d_block_scan(int tid, u32 cnt, u32 *wsum,
     if (lane == 63) wsum[wv] = incl;
     __syncthreads();
     u32 wbase = 0;
+    u32 tot = 0;
 #pragma unroll
     for (int w = 0; w < UA_BLOCK / 64; w++) {
         u32 s = wsum[w];
         if (w < wv) wbase += s;
+        tot += s;
     }
-    total = wsum[0] + wsum[1] + wsum[2] + wsum[3];
+    total = tot;
     excl = wbase + incl - cnt;
 }
 
@@ -545,6 +606,7 @@ __global__ __launch_bounds__(UA_BLOCK) void k_tiles(
     __shared__ u32 scan[UA_BLOCK / 64]; /* per-wave totals for d_block_scan */
     __shared__ u64 s_abefore;
     __shared__ u64 s_bbefore;
+    __shared__ u64 s_run; /* MODE_LOOKBACK: pair-local exclusive prefix */
 
     u64 t = blockIdx.x;
     int tid = threadIdx.x;
@@ -700,19 +762,83 @@ __global__ __launch_bounds__(UA_BLOCK) void k_tiles(
         if (tid == 0) tile_cnt[t] = total;
         return;
     }
+    if (MODE == MODE_LOOKBACK) {
+        /* single-pass: staging = flag array, stage_stride = generation,
+         * offs = per-pair output lengths (d_pout).  See d_lb_resolve. */
+        u64 *lbf = staging;
+        u64 gen = stage_stride;
+        u64 *pout = (u64 *)offs;
+        if (tid == 0) {
+            /* first tile's aggregate IS its inclusive prefix — publish
+             * PREFIX directly so successor windows terminate fast */
+            u64 st = (lt == 0) ? 2ull : 1ull;
+            __hip_atomic_store(&lbf[t], d_lb_word(gen, st, total),
+                               __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+            if (lt == 0) s_run = 0;
+        }
+        if (lt != 0 && tid < 64) {
+            u64 run = d_lb_resolve(lbf, gen, t, d.tile_base, tid);
+            if (tid == 0) {
+                s_run = run;
+                __hip_atomic_store(&lbf[t], d_lb_word(gen, 2ull, run + total),
+                                   __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+            }
+        }
+        __syncthreads();
+        u64 run = s_run;
+        /* out-capacity clamp: on the contract's duplicate-free sorted inputs
+         * emissions never exceed capacity; on invalid inputs results are
+         * unspecified (like the reference's bin variants) but writes stay
+         * in bounds (the reference is memory-safe there, so are we). */
+        u64 cap = (OP == OP_INTERSECT) ? (d.n < d.m ? d.n : d.m)
+                  : (OP == OP_DIFF) ? d.n
+                                    : (d.n + d.m);
+        u64 gbase = run + (u64)excl;
+        if (cnt > 0 && gbase < cap) {
+            u64 *dst = d.out + gbase;
+            u64 room = cap - gbase;
+            int lim = (int)((room < (u64)cnt) ? room : (u64)cnt);
+            int k = 0;
+#pragma unroll
+            for (int s = 0; s < UA_WPT; s++) {
+                if (flags & (1u << s)) {
+                    if (k < lim) dst[k] = em[s];
+                    k++;
+                }
+            }
+        }
+        if (tid == 0) {
+            bool lastt = (t + 1 == total_tiles) || (tile_pair[t + 1] != p);
+            if (lastt) {
+                u64 inc = run + total;
+                pout[p] = inc < cap ? inc : cap;
+            }
+        }
+        return;
+    }
     u64 *dst;
+    u32 lim = (u32)cnt;
     if (MODE == MODE_STAGE) {
+        /* clamp to the staging stride: invalid (duplicate/unsorted) inputs
+         * can emit more than stride entries per tile; results there are
+         * unspecified, OOB writes are not */
+        u32 c0 = excl < (u32)stage_stride ? (u32)stage_stride - excl : 0;
+        if (lim > c0) lim = c0;
         dst = staging + t * stage_stride + excl;
-        if (tid == 0) tile_cnt[t] = total;
+        if (tid == 0)
+            tile_cnt[t] = total < (u32)stage_stride ? total : (u32)stage_stride;
     } else { /* MODE_WRITE: (offs, partials) is the split flat scan */
         dst = d.out +
               (d_off(offs, partials, t) - d_off(offs, partials, d.tile_base)) + excl;
     }
     if (cnt > 0) {
-        int k = 0;
+        u32 k = 0;
 #pragma unroll
         for (int s = 0; s < UA_WPT; s++) {
-            if (flags & (1u << s)) dst[k++] = em[s];
+            if (flags & (1u << s)) {
+                if (k < lim) dst[k] = em[s];
+                k++;
+            }
         }
     }
 }
@@ -1153,7 +1279,7 @@ __global__ __launch_bounds__(UA_BLOCK) void k_compact(
     const UaDesc *__restrict__ descs, const u32 *__restrict__ tile_pair,
     const u32 *__restrict__ tile_cnt, const u64 *__restrict__ offs,
     const u64 *__restrict__ partials, const u64 *__restrict__ staging,
-    u64 stage_stride, u64 total_tiles) {
+    u64 stage_stride, u64 total_tiles, int op) {
     u64 base = ((u64)blockIdx.x * 4 + (threadIdx.x >> 6)) * 4;
     int lane = threadIdx.x & 63;
     if (base >= total_tiles) return;
@@ -1168,7 +1294,16 @@ __global__ __launch_bounds__(UA_BLOCK) void k_compact(
         if (cnt == 0) continue;
         u32 p = tile_pair[t];
         UaDesc d = descs[p];
-        u64 *dst = d.out + (d_off(offs, partials, t) - d_off(offs, partials, d.tile_base));
+        u64 goff = d_off(offs, partials, t) - d_off(offs, partials, d.tile_base);
+        /* pair out-capacity clamp (invalid duplicate/unsorted inputs must
+         * stay memory-safe, like the reference) */
+        u64 cap = (op == OP_INTERSECT) ? (d.n < d.m ? d.n : d.m)
+                  : (op == OP_DIFF) ? d.n
+                                    : (d.n + d.m);
+        if (goff >= cap) continue;
+        u64 room = cap - goff;
+        if ((u64)cnt > room) cnt = (u32)room;
+        u64 *dst = d.out + goff;
         const u64 *src = staging + t * stage_stride;
         for (u32 i = lane; i < cnt; i += 64) dst[i] = src[i];
     }
@@ -1564,7 +1699,7 @@ static thread_local hipError_t g_last_hip = hipSuccess;
 enum {
     WS_DESC = 0, WS_TB, WS_TPAIR, WS_TA0, WS_TCNT, WS_TOFF, WS_PARTIAL,
     WS_STAGE, WS_POUT, WS_HU, WS_HV, WS_HOUT, WS_PACK, WS_SCRATCH_A, WS_SCRATCH_B,
-    WS_COUNT
+    WS_LBF, WS_COUNT
 };
 
 struct ua_ctx {
@@ -1573,6 +1708,10 @@ struct ua_ctx {
     void *ws[WS_COUNT] = {};
     size_t ws_cap[WS_COUNT] = {};
     hipEvent_t ev[4] = {};
+    /* decoupled-lookback flag array (WS_LBF) bookkeeping */
+    uint32_t lbf_gen = 0;
+    size_t lbf_cleared = 0; /* bytes of the CURRENT WS_LBF allocation zeroed */
+    uint32_t *d_viol = nullptr; /* pack-limit validation flag (1 u32) */
     /* stats */
     uint64_t n_launches = 0;
     double kernel_ms = 0.0;
@@ -1591,7 +1730,7 @@ extern "C" const char *ua_strerror(int code) {
     }
 }
 
-extern "C" int ua_version(void) { return 10; }
+extern "C" int ua_version(void) { return 11; }
 
 extern "C" int ua_ctx_create(ua_ctx **out, int device) {
     int ndev = 0;
@@ -1633,6 +1772,31 @@ static int ws_reserve(ua_ctx *c, int slot, size_t bytes) {
         cap = bytes;
     }
     c->ws_cap[slot] = cap;
+    return UA_OK;
+}
+
+#ifndef UA_LOOKBACK
+#define UA_LOOKBACK 1 /* 1 = single-pass decoupled-lookback pipeline (one tile
+                       * kernel per batch run, no scan/compact launches, union
+                       * walks once); 0 = the r01 staged scan+compact path */
+#endif
+
+/* Acquire the ctx-level lookback flag array for T tiles with a fresh
+ * generation.  The array must read as "stale" for the new generation:
+ * guaranteed by zeroing on (re)allocation and re-zeroing when the 16-bit
+ * generation wraps. */
+[[maybe_unused]] static int lb_acquire_ws(ua_ctx *c, u64 T, u64 **flags, u64 *gen) {
+    int rc = ws_reserve(c, WS_LBF, (size_t)T * sizeof(u64));
+    if (rc) return rc;
+    u64 *f = (u64 *)c->ws[WS_LBF];
+    if (c->lbf_cleared < c->ws_cap[WS_LBF] || c->lbf_gen >= UA_LB_GEN_MAX) {
+        HIP_TRY(hipMemsetAsync(f, 0, c->ws_cap[WS_LBF], c->stream));
+        c->lbf_cleared = c->ws_cap[WS_LBF];
+        c->lbf_gen = 0;
+    }
+    c->lbf_gen += 1;
+    *flags = f;
+    *gen = c->lbf_gen;
     return UA_OK;
 }
 
@@ -1695,7 +1859,7 @@ static void launch_tiles(ua_ctx *c, const UaDesc *descs, const u32 *tpair,
                          const u32 *ta0, u64 T, u64 *stage, u64 stride, u32 *tcnt,
                          const u64 *offs, const u64 *part) {
 #if UA_PIPE
-    if constexpr (MODE != MODE_DIRECT) {
+    if constexpr (MODE != MODE_DIRECT && MODE != MODE_LOOKBACK) {
         hipLaunchKernelGGL((k_tiles_pipe<OP, MODE>), dim3((u32)((T + 1) / 2)),
                            dim3(UA_BLOCK), 0, c->stream, descs, tpair, ta0, T, stage,
                            stride, tcnt, offs, part);
@@ -1703,7 +1867,7 @@ static void launch_tiles(ua_ctx *c, const UaDesc *descs, const u32 *tpair,
     }
 #endif
 #if UA_PIPE2
-    {
+    if constexpr (MODE != MODE_LOOKBACK) {
         u64 G = T < 1024 ? T : 1024; /* 4 WGs/CU x 256 CUs resident */
         hipLaunchKernelGGL((k_tiles_p2<OP, MODE>), dim3((u32)G), dim3(UA_BLOCK), 0,
                            c->stream, descs, tpair, ta0, T, stage, stride, tcnt, offs,
@@ -1759,11 +1923,13 @@ static int run_batch_locked(ua_ctx *c, const ua_dpair *pairs, int n_pairs,
     if ((rc = ws_reserve(c, WS_POUT, (size_t)n_pairs * sizeof(u64)))) return rc;
 
     u64 stage_stride = 0;
+#if !UA_LOOKBACK
     if (op == OP_INTERSECT) stage_stride = UA_TILE / 2;
     else if (op == OP_DIFF) stage_stride = UA_TILE;
     if (stage_stride) {
         if ((rc = ws_reserve(c, WS_STAGE, total_tiles * stage_stride * sizeof(u64)))) return rc;
     }
+#endif
 
     UaDesc *d_descs = (UaDesc *)c->ws[WS_DESC];
     u64 *d_tb = (u64 *)((u8 *)c->ws[WS_DESC] + descs_bytes);
@@ -1808,6 +1974,38 @@ static int run_batch_locked(ua_ctx *c, const ua_dpair *pairs, int n_pairs,
             return UA_OK;
         }
 
+#if UA_LOOKBACK
+        u64 *d_lbf;
+        u64 gen;
+        if ((rc = lb_acquire_ws(c, total_tiles, &d_lbf, &gen))) return rc;
+        HIP_TRY(hipEventRecord(c->ev[0], c->stream));
+        if (op == OP_INTERSECT) {
+            launch_tiles<OP_INTERSECT, MODE_LOOKBACK>(c, d_descs, d_tpair, d_ta0,
+                                                      total_tiles, d_lbf, gen, nullptr,
+                                                      d_pout, nullptr);
+        } else if (op == OP_DIFF) {
+            launch_tiles<OP_DIFF, MODE_LOOKBACK>(c, d_descs, d_tpair, d_ta0,
+                                                 total_tiles, d_lbf, gen, nullptr,
+                                                 d_pout, nullptr);
+        } else {
+            launch_tiles<OP_UNION, MODE_LOOKBACK>(c, d_descs, d_tpair, d_ta0,
+                                                  total_tiles, d_lbf, gen, nullptr,
+                                                  d_pout, nullptr);
+        }
+        HIP_TRY(hipEventRecord(c->ev[1], c->stream));
+    }
+    HIP_TRY(hipMemcpyAsync(out_lens, d_pout, (size_t)n_pairs * sizeof(u64),
+                           hipMemcpyDeviceToHost, c->stream));
+    HIP_TRY(hipStreamSynchronize(c->stream));
+    HIP_TRY(hipGetLastError());
+    /* pairs with zero tiles (n+m == 0) never publish a length */
+    for (int p = 0; p < n_pairs; p++)
+        if (pairs[p].n + pairs[p].m == 0) out_lens[p] = 0;
+    (void)d_tcnt;
+    (void)d_toff;
+    (void)d_stage;
+    (void)stage_stride;
+#else
         HIP_TRY(hipEventRecord(c->ev[0], c->stream));
         if (op == OP_INTERSECT) {
             launch_tiles<OP_INTERSECT, MODE_STAGE>(c, d_descs, d_tpair, d_ta0,
@@ -1835,7 +2033,7 @@ static int run_batch_locked(ua_ctx *c, const ua_dpair *pairs, int n_pairs,
             u64 cblk = (total_tiles + 15) / 16;
             hipLaunchKernelGGL(k_compact, dim3((u32)cblk), dim3(UA_BLOCK), 0, c->stream,
                                d_descs, d_tpair, d_tcnt, d_toff, d_part, d_stage,
-                               stage_stride, total_tiles);
+                               stage_stride, total_tiles, op);
         }
     } else {
         if ((rc = ws_reserve(c, WS_PARTIAL, sizeof(u64)))) return rc;
@@ -1850,6 +2048,7 @@ static int run_batch_locked(ua_ctx *c, const ua_dpair *pairs, int n_pairs,
                            hipMemcpyDeviceToHost, c->stream));
     HIP_TRY(hipStreamSynchronize(c->stream));
     HIP_TRY(hipGetLastError());
+#endif
 
     /* stats: HIP-event time of the dominant (tile) kernel(s) on this stream */
     if (total_tiles > 0) {
@@ -1857,12 +2056,14 @@ static int run_batch_locked(ua_ctx *c, const ua_dpair *pairs, int n_pairs,
         HIP_TRY(hipEventElapsedTime(&ms, c->ev[0], c->ev[1]));
         c->kernel_ms += ms;
         c->n_launches += 1;
+#if !UA_LOOKBACK
         if (op == OP_UNION) {
             float ms2 = 0.f;
             HIP_TRY(hipEventElapsedTime(&ms2, c->ev[2], c->ev[3]));
             c->kernel_ms += ms2;
             c->n_launches += 1;
         }
+#endif
     }
     u64 out_elems = 0;
     for (int p = 0; p < n_pairs; p++) out_elems += out_lens[p];
@@ -1897,6 +2098,8 @@ struct ua_batch {
     u64 *d_toff = nullptr;
     u64 *d_part = nullptr;
     u64 *d_pout = nullptr;
+    u64 *d_lbf = nullptr;    /* lookback flag array [total_tiles] */
+    u32 lb_gen = 0;
 };
 
 static size_t align16(size_t x) { return (x + 15) & ~(size_t)15; }
@@ -1932,7 +2135,8 @@ extern "C" int ua_batch_create(ua_ctx *c, const ua_dpair *pairs, int n_pairs,
     size_t o_tpair = align16(o_pout + (size_t)std::max(n_pairs, 1) * sizeof(u64));
     size_t o_ta0 = align16(o_tpair + (T + 1) * sizeof(u32));
     size_t o_tcnt = align16(o_ta0 + (T + 1) * sizeof(u32));
-    size_t total_bytes = align16(o_tcnt + (T + 1) * sizeof(u32));
+    size_t o_lbf = align16(o_tcnt + (T + 1) * sizeof(u32));
+    size_t total_bytes = align16(o_lbf + (T + 1) * sizeof(u64));
     hipError_t e = hipMalloc(&b->mem, total_bytes);
     if (e != hipSuccess) {
         g_last_hip = e;
@@ -1948,6 +2152,7 @@ extern "C" int ua_batch_create(ua_ctx *c, const ua_dpair *pairs, int n_pairs,
     b->d_tpair = (u32 *)(base + o_tpair);
     b->d_ta0 = (u32 *)(base + o_ta0);
     b->d_tcnt = (u32 *)(base + o_tcnt);
+    b->d_lbf = (u64 *)(base + o_lbf);
 
     std::vector<u8> hostbuf(o_tb + tb.size() * sizeof(u64));
     memcpy(hostbuf.data(), descs.data(), descs.size() * sizeof(UaDesc));
@@ -1957,6 +2162,11 @@ extern "C" int ua_batch_create(ua_ctx *c, const ua_dpair *pairs, int n_pairs,
     HIP_TRY(hipMemsetAsync(b->d_tcnt + T, 0, sizeof(u32), c->stream));
     HIP_TRY(hipMemsetAsync(b->d_toff, 0, sizeof(u64), c->stream));
     HIP_TRY(hipMemsetAsync(b->d_part, 0, sizeof(u64), c->stream));
+    /* lookback flags + pout start zeroed (gen 0 never used; zero-tile pairs
+     * never publish a length and stay 0 from here) */
+    HIP_TRY(hipMemsetAsync(b->d_lbf, 0, (T + 1) * sizeof(u64), c->stream));
+    HIP_TRY(hipMemsetAsync(b->d_pout, 0,
+                           (size_t)std::max(n_pairs, 1) * sizeof(u64), c->stream));
     if (T > 0) {
         /* the partition depends only on the (immutable) pair contents:
          * computed once here, reused every run */
@@ -1990,6 +2200,31 @@ extern "C" int ua_batch_run(ua_ctx *c, ua_batch *b, int op, uint64_t *out_lens) 
                                          : -1;
     if (kop < 0) return UA_ERR_INVALID;
     u64 T = b->total_tiles;
+#if UA_LOOKBACK
+    if (T > 0) {
+        if (b->lb_gen >= UA_LB_GEN_MAX) {
+            HIP_TRY(hipMemsetAsync(b->d_lbf, 0, (T + 1) * sizeof(u64), c->stream));
+            b->lb_gen = 0;
+        }
+        b->lb_gen += 1;
+        u64 gen = b->lb_gen;
+        HIP_TRY(hipEventRecord(c->ev[0], c->stream));
+        if (kop == OP_INTERSECT) {
+            launch_tiles<OP_INTERSECT, MODE_LOOKBACK>(c, b->d_descs, b->d_tpair,
+                                                      b->d_ta0, T, b->d_lbf, gen,
+                                                      nullptr, b->d_pout, nullptr);
+        } else if (kop == OP_DIFF) {
+            launch_tiles<OP_DIFF, MODE_LOOKBACK>(c, b->d_descs, b->d_tpair, b->d_ta0,
+                                                 T, b->d_lbf, gen, nullptr, b->d_pout,
+                                                 nullptr);
+        } else {
+            launch_tiles<OP_UNION, MODE_LOOKBACK>(c, b->d_descs, b->d_tpair, b->d_ta0,
+                                                  T, b->d_lbf, gen, nullptr, b->d_pout,
+                                                  nullptr);
+        }
+        HIP_TRY(hipEventRecord(c->ev[1], c->stream));
+    }
+#else
     u64 stride = (kop == OP_INTERSECT) ? UA_TILE / 2 : UA_TILE;
     if (kop != OP_UNION && !b->d_stage) {
         hipError_t e = hipMalloc((void **)&b->d_stage, (T ? T : 1) * UA_TILE * sizeof(u64));
@@ -2027,12 +2262,13 @@ extern "C" int ua_batch_run(ua_ctx *c, ua_batch *b, int op, uint64_t *out_lens) 
         } else {
             hipLaunchKernelGGL(k_compact, dim3((u32)((T + 15) / 16)), dim3(UA_BLOCK), 0,
                                c->stream, b->d_descs, b->d_tpair, b->d_tcnt, b->d_toff,
-                               b->d_part, b->d_stage, stride, T);
+                               b->d_part, b->d_stage, stride, T, kop);
         }
     }
     u64 poutblk = ((u64)b->n_pairs + UA_BLOCK - 1) / UA_BLOCK;
     hipLaunchKernelGGL(k_pair_out, dim3((u32)poutblk), dim3(UA_BLOCK), 0, c->stream,
                        b->d_toff, b->d_part, b->d_tb, b->n_pairs, b->d_pout);
+#endif
     HIP_TRY(hipMemcpyAsync(out_lens, b->d_pout, (size_t)b->n_pairs * sizeof(u64),
                            hipMemcpyDeviceToHost, c->stream));
     HIP_TRY(hipStreamSynchronize(c->stream));
@@ -2043,12 +2279,14 @@ extern "C" int ua_batch_run(ua_ctx *c, ua_batch *b, int op, uint64_t *out_lens) 
         HIP_TRY(hipEventElapsedTime(&ms, c->ev[0], c->ev[1]));
         c->kernel_ms += ms;
         c->n_launches += 1;
+#if !UA_LOOKBACK
         if (kop == OP_UNION) {
             float ms2 = 0.f;
             HIP_TRY(hipEventElapsedTime(&ms2, c->ev[2], c->ev[3]));
             c->kernel_ms += ms2;
             c->n_launches += 1;
         }
+#endif
     }
     u64 out_elems = 0;
     for (int p = 0; p < b->n_pairs; p++) out_elems += out_lens[p];
@@ -2307,11 +2545,19 @@ extern "C" int ua_merge_k_dev(ua_ctx *c, const uint64_t *const *lists,
     if ((rc = ws_reserve(c, WS_TOFF, (max_tiles + 1) * sizeof(u64)))) return rc;
     u64 max_chunks = (max_tiles + 1 + UA_SCAN_CHUNK - 1) / UA_SCAN_CHUNK;
     if ((rc = ws_reserve(c, WS_PARTIAL, (max_chunks + 1) * sizeof(u64)))) return rc;
+#if UA_LOOKBACK
+    /* pre-size the lookback flag array so per-round acquires never realloc
+     * mid-enqueue (hipFree would drain the stream) */
+    if ((rc = ws_reserve(c, WS_LBF, (max_tiles + 1) * sizeof(u64)))) return rc;
+#endif
     u32 *d_tpair = (u32 *)c->ws[WS_TPAIR];
     u32 *d_ta0 = (u32 *)c->ws[WS_TA0];
     u32 *d_tcnt = (u32 *)c->ws[WS_TCNT];
     u64 *d_toff = (u64 *)c->ws[WS_TOFF];
     u64 *d_part = (u64 *)c->ws[WS_PARTIAL];
+    (void)d_tcnt;
+    (void)d_toff;
+    (void)d_part;
 
     HIP_TRY(hipEventRecord(c->ev[0], c->stream));
     int lcur = 0;
@@ -2323,6 +2569,24 @@ extern "C" int ua_merge_k_dev(ua_ctx *c, const uint64_t *const *lists,
         u64 mk = ((u64)rd.npair + UA_BLOCK - 1) / UA_BLOCK;
         hipLaunchKernelGGL(k_make_descs, dim3((u32)mk), dim3(UA_BLOCK), 0, c->stream,
                            d_descs, d_lens[lcur], rd.nk_prev, rd.npair);
+#if UA_LOOKBACK
+        /* zero-capacity pairs have no tiles and never publish a length */
+        HIP_TRY(hipMemsetAsync(d_lens[lcur ^ 1], 0, (size_t)rd.npair * sizeof(u64),
+                               c->stream));
+        if (T > 0) {
+            u64 pblk = (T + UA_BLOCK - 1) / UA_BLOCK;
+            hipLaunchKernelGGL(k_partition, dim3((u32)pblk), dim3(UA_BLOCK), 0, c->stream,
+                               d_descs, d_tb, rd.npair, T, d_tpair, d_ta0, 0);
+            hipLaunchKernelGGL(k_partition, dim3((u32)pblk), dim3(UA_BLOCK), 0, c->stream,
+                               d_descs, d_tb, rd.npair, T, d_tpair, d_ta0, 1);
+            u64 *d_lbf;
+            u64 gen;
+            if ((rc = lb_acquire_ws(c, T, &d_lbf, &gen))) return rc;
+            launch_tiles<OP_UNION, MODE_LOOKBACK>(c, d_descs, d_tpair, d_ta0, T,
+                                                  d_lbf, gen, nullptr,
+                                                  d_lens[lcur ^ 1], nullptr);
+        }
+#else
         HIP_TRY(hipMemsetAsync(d_tcnt + T, 0, sizeof(u32), c->stream));
         if (T > 0) {
             u64 pblk = (T + UA_BLOCK - 1) / UA_BLOCK;
@@ -2346,6 +2610,7 @@ extern "C" int ua_merge_k_dev(ua_ctx *c, const uint64_t *const *lists,
         u64 poutblk = ((u64)rd.npair + UA_BLOCK - 1) / UA_BLOCK;
         hipLaunchKernelGGL(k_pair_out, dim3((u32)poutblk), dim3(UA_BLOCK), 0, c->stream,
                            d_toff, d_part, d_tb, rd.npair, d_lens[lcur ^ 1]);
+#endif
         lcur ^= 1;
     }
     /* final item: data in ptrs[0] (capacity position), length in d_lens[lcur][0] */
@@ -2363,10 +2628,56 @@ extern "C" int ua_merge_k_dev(ua_ctx *c, const uint64_t *const *lists,
     HIP_TRY(hipStreamSynchronize(c->stream));
     HIP_TRY(hipEventElapsedTime(&ms, c->ev[0], c->ev[1]));
     c->kernel_ms += ms;
+#if UA_LOOKBACK
+    c->n_launches += R;
+    c->bytes_algo += 8 * 2 * cap_work; /* capacity upper bound (in + out, one pass) */
+#else
     c->n_launches += 2 * R;
     c->bytes_algo += 8 * 3 * cap_work; /* capacity upper bound (union count+write) */
+#endif
     *out_n = final_len;
     return UA_OK;
+}
+
+/* Pack-limit validation: k_packed decodes blocks only up to
+ * num_uids <= UA_MAX_BLOCK_UIDS and deltas_len <= UA_MAX_DELTAS.  The
+ * reference supports arbitrary BlockSize, so packs encoded elsewhere must be
+ * REJECTED (UA_ERR_INVALID), not silently zeroed by the in-kernel guard
+ * (ADVICE r01). */
+__global__ __launch_bounds__(UA_BLOCK) void k_validate_pack(
+    const u32 *__restrict__ num_uids, const u64 *__restrict__ delta_offs, u64 nb,
+    u32 *__restrict__ viol) {
+    u64 b = (u64)blockIdx.x * UA_BLOCK + threadIdx.x;
+    if (b >= nb) return;
+    if (num_uids[b] > UA_MAX_BLOCK_UIDS ||
+        delta_offs[b + 1] - delta_offs[b] > UA_MAX_DELTAS)
+        atomicOr(viol, 1u);
+}
+
+/* enqueue the validation on the ctx stream (read the flag after a sync) */
+static int pack_validate_begin(ua_ctx *c, const u32 *num_uids, const u64 *delta_offs,
+                               u64 nb) {
+    if (!c->d_viol) {
+        hipError_t e = hipMalloc((void **)&c->d_viol, sizeof(u32));
+        if (e != hipSuccess) {
+            g_last_hip = e;
+            return UA_ERR_NOMEM;
+        }
+    }
+    HIP_TRY(hipMemsetAsync(c->d_viol, 0, sizeof(u32), c->stream));
+    if (nb) {
+        u64 g = (nb + UA_BLOCK - 1) / UA_BLOCK;
+        hipLaunchKernelGGL(k_validate_pack, dim3((u32)g), dim3(UA_BLOCK), 0, c->stream,
+                           num_uids, delta_offs, nb, c->d_viol);
+    }
+    return UA_OK;
+}
+
+/* read the flag; stream must already be synchronized */
+static int pack_validate_end(ua_ctx *c) {
+    u32 viol = 0;
+    HIP_TRY(hipMemcpy(&viol, c->d_viol, sizeof(u32), hipMemcpyDeviceToHost));
+    return viol ? UA_ERR_INVALID : UA_OK;
 }
 
 /* ---- packed pipeline ---- */
@@ -2387,6 +2698,7 @@ static int run_packed_locked(ua_ctx *c, const ua_dpack *pk, u64 after, const u64
     u64 *d_stage = (u64 *)c->ws[WS_STAGE];
     HIP_TRY(hipMemsetAsync(d_cnt + nb, 0, sizeof(u32), c->stream));
 
+    if ((rc = pack_validate_begin(c, pk->num_uids, pk->delta_offs, nb))) return rc;
     u64 nwg = (nb + UA_PKW - 1) / UA_PKW;
     HIP_TRY(hipEventRecord(c->ev[0], c->stream));
     if (decode_only) {
@@ -2426,6 +2738,10 @@ static int run_packed_locked(ua_ctx *c, const ua_dpack *pk, u64 after, const u64
     HIP_TRY(hipMemcpyAsync(&dbytes, pk->delta_offs + nb, sizeof(u64),
                            hipMemcpyDeviceToHost, c->stream));
     HIP_TRY(hipStreamSynchronize(c->stream));
+    if ((rc = pack_validate_end(c))) {
+        *out_n = 0;
+        return rc;
+    }
     c->bytes_algo += hdr + dbytes + 8 * (m + *out_n);
     return UA_OK;
 }
@@ -2474,6 +2790,7 @@ extern "C" int ua_intersect_packed_batch_dev(ua_ctx *c, const uint64_t *bases,
     HIP_TRY(hipMemsetAsync(d_cnt + nb, 0, sizeof(u32), c->stream));
 
     if (nb > 0) {
+        if ((rc = pack_validate_begin(c, num_uids, delta_offs, nb))) return rc;
         u64 nwg = (nb + UA_PKW - 1) / UA_PKW;
         HIP_TRY(hipEventRecord(c->ev[0], c->stream));
         hipLaunchKernelGGL(k_packed<0>, dim3((u32)nwg), dim3(UA_BLOCK), 0, c->stream,
@@ -2499,6 +2816,10 @@ extern "C" int ua_intersect_packed_batch_dev(ua_ctx *c, const uint64_t *bases,
                            hipMemcpyDeviceToHost, c->stream));
     HIP_TRY(hipStreamSynchronize(c->stream));
     HIP_TRY(hipGetLastError());
+    if (nb > 0 && (rc = pack_validate_end(c))) {
+        for (int p = 0; p < n_packs; p++) out_lens[p] = 0;
+        return rc;
+    }
 
     if (nb > 0) {
         float ms = 0.f;
@@ -2592,7 +2913,13 @@ extern "C" int ua_pbatch_create(ua_ctx *c, const uint64_t *bases, const uint32_t
     u64 db = 0;
     HIP_TRY(hipMemcpyAsync(&db, delta_offs + b->nb, sizeof(u64),
                            hipMemcpyDeviceToHost, c->stream));
+    int vrc = (b->nb > 0) ? pack_validate_begin(c, num_uids, delta_offs, b->nb) : UA_OK;
     HIP_TRY(hipStreamSynchronize(c->stream));
+    if (vrc == UA_OK && b->nb > 0) vrc = pack_validate_end(c);
+    if (vrc != UA_OK) {
+        ua_pbatch_destroy(c, b);
+        return vrc;
+    }
     b->deltas_bytes = db;
     *out = b;
     return UA_OK;
