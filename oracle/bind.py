@@ -93,7 +93,7 @@ def lib():
             C.POINTER(_Pack), C.c_uint64, _u64p, C.c_size_t, _u64p]
         L.orc_intersect_batch_cpu.argtypes = [
             C.c_int, C.POINTER(_u64p), _szp, C.POINTER(_u64p), _szp,
-            C.POINTER(_u64p), _szp, C.c_int]
+            C.POINTER(_u64p), _szp, C.c_int, C.POINTER(C.c_int)]
         L.orc_omp_max_threads.restype = C.c_int
         _lib = L
     return _lib
@@ -265,7 +265,7 @@ def intersect_compressed_with(pack: Pack, after, v):
     return out[:n].copy()
 
 
-def intersect_batch_cpu(us, vs, n_threads=0):
+def intersect_batch_cpu(us, vs, n_threads=0, return_threads_used=False):
     """OpenMP batched IntersectWith across pairs — the CPU baseline leg."""
     us = [_arr(u) for u in us]
     vs = [_arr(v) for v in vs]
@@ -277,8 +277,12 @@ def intersect_batch_cpu(us, vs, n_threads=0):
     un = (C.c_size_t * k)(*[a.size for a in us])
     vn = (C.c_size_t * k)(*[a.size for a in vs])
     on = (C.c_size_t * k)()
-    lib().orc_intersect_batch_cpu(k, up, un, vp, vn, op, on, n_threads)
-    return [outs[i][:on[i]].copy() for i in range(k)]
+    tu = C.c_int(0)
+    lib().orc_intersect_batch_cpu(k, up, un, vp, vn, op, on, n_threads, C.byref(tu))
+    res = [outs[i][:on[i]].copy() for i in range(k)]
+    if return_threads_used:
+        return res, tu.value
+    return res
 
 
 def omp_max_threads():

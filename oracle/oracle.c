@@ -767,16 +767,32 @@ int orc_omp_max_threads(void) {
 #endif
 }
 
+/* threads_used (optional): how many distinct OpenMP threads actually ran
+ * >=1 pair — evidence that the baseline really used the cores it claims
+ * (bench.py reports it as cpu_baseline.cores). */
 void orc_intersect_batch_cpu(int n_pairs,
                              const uint64_t *const *us, const size_t *ns,
                              const uint64_t *const *vs, const size_t *ms,
                              uint64_t *const *outs, size_t *out_ns,
-                             int n_threads) {
+                             int n_threads, int *threads_used) {
+    static unsigned char used[4096];
+    memset(used, 0, sizeof(used));
 #ifdef _OPENMP
     if (n_threads > 0) omp_set_num_threads(n_threads);
 #pragma omp parallel for schedule(dynamic)
 #endif
     for (int p = 0; p < n_pairs; p++) {
+#ifdef _OPENMP
+        int tid = omp_get_thread_num();
+#else
+        int tid = 0;
+#endif
+        if (tid >= 0 && tid < (int)sizeof(used)) used[tid] = 1;
         out_ns[p] = orc_intersect_with(us[p], ns[p], vs[p], ms[p], outs[p]);
+    }
+    if (threads_used) {
+        int tu = 0;
+        for (size_t i = 0; i < sizeof(used); i++) tu += used[i];
+        *threads_used = tu;
     }
 }

@@ -131,12 +131,13 @@ void orc_intersect_compressed_with_lin_jump(orc_dec *dec, const uint64_t *v, siz
 void orc_intersect_compressed_with_bin(orc_dec *dec, const uint64_t *q, size_t lq,
                                        uint64_t *o, size_t *o_n); /* :87 */
 
-/* ---- batched CPU baseline (OpenMP across pairs; bench.py cpu_baseline leg) ---- */
+/* ---- batched CPU baseline (OpenMP across pairs; bench.py cpu_baseline leg) ----
+ * threads_used (nullable): distinct threads that ran >=1 pair. */
 void orc_intersect_batch_cpu(int n_pairs,
                              const uint64_t *const *us, const size_t *ns,
                              const uint64_t *const *vs, const size_t *ms,
                              uint64_t *const *outs, size_t *out_ns,
-                             int n_threads);
+                             int n_threads, int *threads_used);
 int orc_omp_max_threads(void);
 
 #ifdef __cplusplus
