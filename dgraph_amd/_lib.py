@@ -42,6 +42,15 @@ class UaDPack(C.Structure):
     ]
 
 
+class UaDFilter(C.Structure):
+    _fields_ = [
+        ("u", C.c_void_p),
+        ("n", _u64),
+        ("mask", C.c_void_p),
+        ("out", C.c_void_p),
+    ]
+
+
 class UaDSeg(C.Structure):
     _fields_ = [
         ("data", C.c_void_p),
@@ -111,6 +120,9 @@ def lib():
         L.ua_sort_segments_dev.argtypes = [C.c_void_p, C.POINTER(UaDSeg), C.c_int]
         L.ua_index_of_batch_dev.argtypes = [C.c_void_p, C.c_void_p, _u64, C.c_void_p, _u64,
                                             C.c_void_p]
+        L.ua_apply_filter_batch_dev.argtypes = [C.c_void_p, C.POINTER(UaDFilter),
+                                                C.c_int, _u64p]
+        L.ua_apply_filter.argtypes = [C.c_void_p, _u64p, _u64, _u8p, _u64p]
         L.ua_intersect_k_dev.argtypes = [C.c_void_p, _voidpp, _u64p, C.c_int, C.c_void_p, _u64p]
         L.ua_merge_k_dev.argtypes = [C.c_void_p, _voidpp, _u64p, C.c_int, C.c_void_p, _u64p]
         L.ua_intersect_packed_dev.argtypes = [C.c_void_p, C.POINTER(UaDPack), _u64, C.c_void_p,

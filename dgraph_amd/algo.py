@@ -187,6 +187,30 @@ class Engine:
         check(lib().ua_sort_segments_dev(self._ctx, segs, k))
         return tensors
 
+    def apply_filter_batch(self, us, masks, outs=None):
+        """Batched algo.ApplyFilter (uidlist.go:21): order-preserving mask
+        compaction on device.  us: CUDA int64 tensors; masks: CUDA uint8/bool
+        tensors (nonzero = keep).  outs may be the us themselves (in-place,
+        the reference's shape).  Returns (outs, lens)."""
+        import torch
+        from dgraph_amd._lib import UaDFilter
+        k = len(us)
+        if outs is None:
+            outs = us  # in-place like the reference
+        tasks = (UaDFilter * max(k, 1))()
+        for i in range(k):
+            m = masks[i]
+            if m.dtype == torch.bool:
+                m = m.view(torch.uint8)
+            assert m.dtype == torch.uint8 and m.numel() == us[i].numel()
+            tasks[i].u = us[i].data_ptr()
+            tasks[i].n = us[i].numel()
+            tasks[i].mask = m.data_ptr()
+            tasks[i].out = outs[i].data_ptr()
+        lens = (_u64 * max(k, 1))()
+        check(lib().ua_apply_filter_batch_dev(self._ctx, tasks, k, lens))
+        return outs, [int(lens[i]) for i in range(k)]
+
     def index_of_batch(self, u, queries):
         """Batched algo.IndexOf: u, queries CUDA int64; returns int64 tensor
         of positions (-1 = absent)."""
@@ -534,13 +558,29 @@ def index_of(u, uid):
     return int(lib().ua_index_of(_hptr(u), u.size, _u64(uid)))
 
 
-def apply_filter(u, mask):
+def apply_filter(u, mask, engine=None):
     """algo.ApplyFilter (uidlist.go:21): boolean-mask compaction.
 
     The reference takes a Go closure; across a C ABI the filter arrives as a
-    precomputed mask (the callers evaluate per-uid predicates upstream)."""
+    precomputed mask (the callers evaluate per-uid predicates upstream).
+    With an Engine, host arrays go through the ua_apply_filter C-ABI export
+    (in-place device compaction — the cgo drop-in path); CUDA tensors go
+    through the batched device export."""
     import torch
     if isinstance(u, torch.Tensor):
+        if engine is not None and u.is_cuda:
+            m = mask if isinstance(mask, torch.Tensor) else \
+                torch.as_tensor(np.asarray(mask, dtype=np.uint8), device=u.device)
+            out = torch.empty_like(u)
+            outs, lens = engine.apply_filter_batch([u], [m], [out])
+            return out[:lens[0]]
         return u[mask]
-    u = _np_u64(u)
-    return u[np.asarray(mask, dtype=bool)]
+    u = np.ascontiguousarray(_np_u64(u))
+    m = np.ascontiguousarray(np.asarray(mask, dtype=bool).view(np.uint8))
+    if engine is not None:
+        from dgraph_amd._lib import _u8p
+        n = _u64()
+        check(lib().ua_apply_filter(engine._ctx, _hptr(u), u.size,
+                                    m.ctypes.data_as(_u8p), C.byref(n)))
+        return u[:n.value]
+    return u[m.view(bool)]

@@ -1330,6 +1330,95 @@ __global__ __launch_bounds__(UA_BLOCK) void k_compact_flat(
     }
 }
 
+/* ==================== kernel: batched ApplyFilter ==================== */
+
+/* algo.ApplyFilter (uidlist.go:21): in-place mask compaction.  The Go
+ * closure f(uid, i) becomes a precomputed per-element byte mask across the
+ * C-ABI (callers evaluate predicates upstream, worker/task.go:1403).  One
+ * workgroup per 2048-element chunk; per-task output offsets by the same
+ * decoupled-lookback protocol as k_tiles (MODE_LOOKBACK).  out == u
+ * (in-place, the reference's shape) is safe: every chunk's loads land
+ * (vmcnt) before its count is published, and successors only write after
+ * all predecessors published. */
+struct alignas(16) UaFDesc {
+    const u64 *u;
+    u64 n;
+    const u8 *mask;
+    u64 *out;
+    u64 tile_base;
+};
+
+__global__ __launch_bounds__(UA_BLOCK) void k_filter(
+    const UaFDesc *__restrict__ descs, const u64 *__restrict__ tb, int n_tasks,
+    u64 total_tiles, u64 *__restrict__ lbf, u64 gen, u64 *__restrict__ pout) {
+    __shared__ u32 scan[UA_BLOCK / 64];
+    __shared__ u64 s_run;
+    u64 t = blockIdx.x;
+    int tid = threadIdx.x;
+    /* task p with tb[p] <= t < tb[p+1] */
+    int lo = 0, hi = n_tasks - 1;
+    while (lo < hi) {
+        int mid = (lo + hi + 1) >> 1;
+        if (tb[mid] <= t) lo = mid;
+        else hi = mid - 1;
+    }
+    int p = lo;
+    UaFDesc d = descs[p];
+    u64 lt = t - d.tile_base;
+    u64 d0 = lt * UA_TILE;
+    u64 d1 = d0 + UA_TILE;
+    if (d1 > d.n) d1 = d.n;
+    /* per-thread contiguous 8-element run (order-preserving compaction;
+     * the 64-B-strided lane pattern coalesces through L1 over the 8 steps) */
+    u64 base = d0 + (u64)tid * UA_WPT;
+    u64 vals[UA_WPT];
+    u32 flags = 0;
+    int cnt = 0;
+#pragma unroll
+    for (int s = 0; s < UA_WPT; s++) {
+        u64 i = base + s;
+        if (i < d1) {
+            vals[s] = d.u[i];
+            u32 keep = d.mask[i] ? 1u : 0u;
+            flags |= keep << s;
+            cnt += (int)keep;
+        }
+    }
+    /* in-place safety: loads must LAND chip-wide before this chunk's count
+     * is published (a successor may overwrite these addresses) */
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    u32 excl, total;
+    d_block_scan(tid, (u32)cnt, scan, excl, total);
+    if (tid == 0) {
+        u64 st = (lt == 0) ? 2ull : 1ull;
+        __hip_atomic_store(&lbf[t], d_lb_word(gen, st, total), __ATOMIC_RELAXED,
+                           __HIP_MEMORY_SCOPE_AGENT);
+        if (lt == 0) s_run = 0;
+    }
+    if (lt != 0 && tid < 64) {
+        u64 run = d_lb_resolve(lbf, gen, t, d.tile_base, tid);
+        if (tid == 0) {
+            s_run = run;
+            __hip_atomic_store(&lbf[t], d_lb_word(gen, 2ull, run + total),
+                               __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+        }
+    }
+    __syncthreads();
+    u64 run = s_run;
+    if (cnt > 0) {
+        u64 *dst = d.out + run + excl;
+        int k = 0;
+#pragma unroll
+        for (int s = 0; s < UA_WPT; s++) {
+            if (flags & (1u << s)) dst[k++] = vals[s];
+        }
+    }
+    if (tid == 0) {
+        bool lastt = (lt + 1) * UA_TILE >= d.n;
+        if (lastt) pout[p] = run + total;
+    }
+}
+
 /* ==================== kernel: batched IndexOf ==================== */
 
 __global__ __launch_bounds__(UA_BLOCK) void k_index_of(const u64 *__restrict__ u, u64 n,
@@ -2379,6 +2468,91 @@ extern "C" int ua_sort_segments_dev(ua_ctx *c, const ua_dseg *segs, int n_segs) 
         width <<= 1;
     }
     return UA_OK; /* every segment ends in data by start-side parity */
+}
+
+/* ---- batched algo.ApplyFilter (uidlist.go:21; callers worker/task.go:1403,
+ * query/query.go:1431) ---- */
+extern "C" int ua_apply_filter_batch_dev(ua_ctx *c, const ua_dfilter *tasks,
+                                         int n_tasks, uint64_t *out_lens) {
+    std::lock_guard<std::recursive_mutex> g(c->mu);
+    if (n_tasks <= 0) return UA_OK;
+    HIP_TRY(hipSetDevice(c->device));
+    std::vector<UaFDesc> descs((size_t)n_tasks);
+    std::vector<u64> tb((size_t)n_tasks + 1);
+    u64 T = 0, in_bytes = 0;
+    for (int p = 0; p < n_tasks; p++) {
+        const ua_dfilter &f = tasks[p];
+        if (f.n >= (1ull << 31)) return UA_ERR_INVALID;
+        descs[p] = {f.u, f.n, f.mask, f.out, T};
+        tb[p] = T;
+        T += (f.n + UA_TILE - 1) / UA_TILE;
+        in_bytes += 9 * f.n; /* 8 B uid + 1 B mask */
+    }
+    tb[n_tasks] = T;
+    int rc;
+    size_t descs_bytes = descs.size() * sizeof(UaFDesc);
+    size_t tb_bytes = tb.size() * sizeof(u64);
+    if ((rc = ws_reserve(c, WS_DESC, descs_bytes + tb_bytes))) return rc;
+    if ((rc = ws_reserve(c, WS_POUT, (size_t)n_tasks * sizeof(u64)))) return rc;
+    UaFDesc *d_descs = (UaFDesc *)c->ws[WS_DESC];
+    u64 *d_tb = (u64 *)((u8 *)c->ws[WS_DESC] + descs_bytes);
+    u64 *d_pout = (u64 *)c->ws[WS_POUT];
+    std::vector<u8> hostbuf(descs_bytes + tb_bytes);
+    memcpy(hostbuf.data(), descs.data(), descs_bytes);
+    memcpy(hostbuf.data() + descs_bytes, tb.data(), tb_bytes);
+    HIP_TRY(hipMemcpyAsync(d_descs, hostbuf.data(), hostbuf.size(),
+                           hipMemcpyHostToDevice, c->stream));
+    if (T > 0) {
+        u64 *d_lbf;
+        u64 gen;
+        if ((rc = lb_acquire_ws(c, T, &d_lbf, &gen))) return rc;
+        HIP_TRY(hipEventRecord(c->ev[0], c->stream));
+        hipLaunchKernelGGL(k_filter, dim3((u32)T), dim3(UA_BLOCK), 0, c->stream,
+                           d_descs, d_tb, n_tasks, T, d_lbf, gen, d_pout);
+        HIP_TRY(hipEventRecord(c->ev[1], c->stream));
+    }
+    HIP_TRY(hipMemcpyAsync(out_lens, d_pout, (size_t)n_tasks * sizeof(u64),
+                           hipMemcpyDeviceToHost, c->stream));
+    HIP_TRY(hipStreamSynchronize(c->stream));
+    HIP_TRY(hipGetLastError());
+    u64 out_elems = 0;
+    for (int p = 0; p < n_tasks; p++) {
+        if (tasks[p].n == 0) out_lens[p] = 0; /* no tiles -> never published */
+        out_elems += out_lens[p];
+    }
+    if (T > 0) {
+        float ms = 0.f;
+        HIP_TRY(hipEventElapsedTime(&ms, c->ev[0], c->ev[1]));
+        c->kernel_ms += ms;
+        c->n_launches += 1;
+    }
+    c->bytes_algo += in_bytes + 8 * out_elems;
+    return UA_OK;
+}
+
+/* host-pointer algo.ApplyFilter: in-place like the reference (u is compacted
+ * to *out_n kept elements); upload + device compaction + download */
+extern "C" int ua_apply_filter(ua_ctx *c, uint64_t *u, uint64_t n,
+                               const uint8_t *mask, uint64_t *out_n) {
+    std::lock_guard<std::recursive_mutex> g(c->mu);
+    HIP_TRY(hipSetDevice(c->device));
+    if (n == 0) {
+        *out_n = 0;
+        return UA_OK;
+    }
+    int rc;
+    if ((rc = ws_reserve(c, WS_HU, n * sizeof(u64)))) return rc;
+    if ((rc = ws_reserve(c, WS_HV, n))) return rc;
+    u64 *d_u = (u64 *)c->ws[WS_HU];
+    u8 *d_mask = (u8 *)c->ws[WS_HV];
+    HIP_TRY(hipMemcpyAsync(d_u, u, n * sizeof(u64), hipMemcpyHostToDevice, c->stream));
+    HIP_TRY(hipMemcpyAsync(d_mask, mask, n, hipMemcpyHostToDevice, c->stream));
+    ua_dfilter task = {d_u, n, d_mask, d_u}; /* in-place on device too */
+    if ((rc = ua_apply_filter_batch_dev(c, &task, 1, out_n))) return rc;
+    HIP_TRY(hipMemcpyAsync(u, d_u, *out_n * sizeof(u64), hipMemcpyDeviceToHost,
+                           c->stream));
+    HIP_TRY(hipStreamSynchronize(c->stream));
+    return UA_OK;
 }
 
 extern "C" int ua_index_of_batch_dev(ua_ctx *c, const uint64_t *u, uint64_t n,

@@ -121,7 +121,11 @@ int ua_encode_dev(ua_ctx *, const uint64_t *uids, uint64_t n,
  * All pointers in ua_dpair are DEVICE pointers; the descriptor array itself
  * and out_lens live on the host.  Inputs are sorted uint64 lists; results are
  * bit-exact vs the reference on duplicate-free inputs (the parity domain
- * pinned by uidlist_test.go:394,536-542 — see DESIGN.md §semantics). */
+ * pinned by uidlist_test.go:394,536-542 — see DESIGN.md §semantics).
+ * On inputs that violate the contract (duplicates / unsorted), results are
+ * unspecified — like the reference's bin variants — but all writes stay
+ * within each pair's out capacity (the engine clamps; the reference is
+ * memory-safe on such inputs and so is this ABI). */
 typedef struct {
     const uint64_t *u; /* device */
     uint64_t n;
@@ -184,6 +188,20 @@ int ua_merge_k_dev(ua_ctx *, const uint64_t *const *lists,
 int ua_index_of_batch_dev(ua_ctx *, const uint64_t *u, uint64_t n,
                           const uint64_t *queries, uint64_t nq, int64_t *out);
 
+/* batched algo.ApplyFilter (uidlist.go:21; callers worker/task.go:1403,
+ * query/query.go:1431): order-preserving mask compaction.  The reference
+ * takes a Go closure f(uid, i); across the C-ABI the filter arrives as a
+ * precomputed per-element byte mask (callers evaluate per-uid predicates
+ * upstream).  out may equal u (in-place, the reference's shape). */
+typedef struct {
+    const uint64_t *u;    /* device */
+    uint64_t n;
+    const uint8_t *mask;  /* device, n bytes; nonzero = keep */
+    uint64_t *out;        /* device, capacity >= n; may alias u */
+} ua_dfilter;
+int ua_apply_filter_batch_dev(ua_ctx *, const ua_dfilter *tasks, int n_tasks,
+                              uint64_t *out_lens);
+
 /* fused decode+intersect: algo.IntersectCompressedWith (uidlist.go:33) over a
  * device pack; v device; out device, capacity min(total_uids, m). */
 int ua_intersect_packed_dev(ua_ctx *, const ua_dpack *, uint64_t after_uid,
@@ -238,6 +256,8 @@ int ua_difference(ua_ctx *, const uint64_t *u, uint64_t n, const uint64_t *v,
 int64_t ua_index_of(const uint64_t *u, uint64_t n, uint64_t uid); /* IndexOf:
                   * one log(n) probe — host binary search like the reference;
                   * the batched GPU form is ua_index_of_batch_dev */
+int ua_apply_filter(ua_ctx *, uint64_t *u /* compacted in place */, uint64_t n,
+                    const uint8_t *mask, uint64_t *out_n); /* ApplyFilter */
 int ua_intersect_packed(ua_ctx *, const ua_pack *, uint64_t after_uid,
                         const uint64_t *v, uint64_t m, uint64_t *out,
                         uint64_t *out_n);            /* IntersectCompressedWith */

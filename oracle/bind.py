@@ -88,6 +88,8 @@ def lib():
         L.orc_difference.argtypes = [_u64p, C.c_size_t, _u64p, C.c_size_t, _u64p]
         L.orc_index_of.restype = C.c_int64
         L.orc_index_of.argtypes = [_u64p, C.c_size_t, C.c_uint64]
+        L.orc_apply_filter.restype = C.c_size_t
+        L.orc_apply_filter.argtypes = [_u64p, C.c_size_t, C.POINTER(C.c_uint8)]
         L.orc_intersect_compressed_with.restype = C.c_size_t
         L.orc_intersect_compressed_with.argtypes = [
             C.POINTER(_Pack), C.c_uint64, _u64p, C.c_size_t, _u64p]
@@ -157,6 +159,16 @@ def merge_sorted(lists):
 def index_of(u, uid):
     u = _arr(u)
     return int(lib().orc_index_of(_ptr(u), u.size, C.c_uint64(uid)))
+
+
+def apply_filter(u, mask):
+    """uidlist.go:21 ApplyFilter (mask = precomputed f(uid, i))."""
+    u = _arr(u).copy()  # in-place in C; keep the caller's array intact
+    m = np.ascontiguousarray(np.asarray(mask, dtype=bool).view(np.uint8))
+    assert m.size == u.size
+    n = lib().orc_apply_filter(_ptr(u), u.size,
+                               m.ctypes.data_as(C.POINTER(C.c_uint8)))
+    return u[:n].copy()
 
 
 class Pack:

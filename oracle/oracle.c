@@ -666,6 +666,16 @@ size_t orc_difference(const uint64_t *u, size_t n, const uint64_t *v, size_t m,
     return o;
 }
 
+/* uidlist.go:21 ApplyFilter: in-place keep-where-f compaction; the Go
+ * closure f(uid, i) is a precomputed mask here (the C-ABI's contract). */
+size_t orc_apply_filter(uint64_t *u, size_t n, const uint8_t *mask) {
+    size_t o = 0;
+    for (size_t i = 0; i < n; i++) {
+        if (mask[i]) u[o++] = u[i];
+    }
+    return o;
+}
+
 /* uidlist.go:546 IndexOf */
 int64_t orc_index_of(const uint64_t *u, size_t n, uint64_t uid) {
     size_t i = lower_bound_u64(u, n, uid);

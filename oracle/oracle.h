@@ -121,6 +121,8 @@ size_t orc_difference(const uint64_t *u, size_t n, const uint64_t *v, size_t m,
 
 /* IndexOf  :546 */
 int64_t orc_index_of(const uint64_t *u, size_t n, uint64_t uid);
+/* ApplyFilter  :21 (in-place; mask = precomputed f(uid,i)) */
+size_t orc_apply_filter(uint64_t *u, size_t n, const uint8_t *mask);
 
 /* IntersectCompressedWith dispatch (linVsBinRatio=10)  :33 */
 size_t orc_intersect_compressed_with(const orc_pack *pack, uint64_t after_uid,

@@ -82,3 +82,15 @@ def test_apply_filter_mask():
     u = np.array([1, 2, 3, 4, 5], dtype=np.uint64)
     got = algo.apply_filter(u, (u % 2) == 1)
     assert got.tolist() == [1, 3, 5]
+
+
+def test_oracle_apply_filter():
+    """orc_apply_filter (uidlist.go:21) vs numpy + the golden table shape."""
+    rng = np.random.default_rng(0xD6A77)
+    for n in [0, 1, 5, 1000, 4097]:
+        u = np.sort(rng.integers(0, 1 << 48, size=n, dtype=np.uint64))
+        mask = rng.integers(0, 2, size=n, dtype=np.uint8).astype(bool)
+        got = orc.apply_filter(u, mask)
+        assert got.tolist() == u[mask].tolist()
+    u = np.array([1, 2, 3, 4, 5], dtype=np.uint64)
+    assert orc.apply_filter(u, (u % 2) == 1).tolist() == [1, 3, 5]

@@ -572,3 +572,51 @@ def test_host_packed_api(eng):
             orc.intersect_compressed_with(opack, 0, v).tolist()
     finally:
         _lib.lib().ua_owned_pack_free(h)
+
+
+# ---------- ApplyFilter across the C-ABI (uidlist.go:21; worker/task.go:1403) ----------
+
+@pytest.mark.parametrize("n", [0, 1, 7, 2048, 2049, 100_000])
+def test_apply_filter_host_abi(eng, n):
+    """ua_apply_filter (host-pointer, in-place like the reference) vs oracle."""
+    rng = np.random.default_rng(SEED + n)
+    u = np.sort(rng.choice(np.uint64(1) << np.uint64(48), size=n, replace=False)) \
+        if n else np.empty(0, dtype=np.uint64)
+    u = u.astype(np.uint64)
+    mask = rng.integers(0, 2, size=n, dtype=np.uint8).astype(bool)
+    got = algo.apply_filter(u.copy(), mask, engine=eng)
+    want = orc.apply_filter(u, mask)
+    assert got.tolist() == want.tolist()
+
+
+@pytest.mark.parametrize("inplace", [True, False])
+def test_apply_filter_batch_dev(eng, inplace):
+    """Batched device mask compaction, mixed sizes incl. tile boundaries;
+    in-place (out == u, the reference's shape) and separate-out forms."""
+    rng = np.random.default_rng(SEED)
+    sizes = [0, 1, 5, 2048, 2049, 4096, 40_000, 1_000_000]
+    us_np, masks_np = [], []
+    for i, n in enumerate(sizes):
+        u = np.sort(rng.choice(np.uint64(1) << np.uint64(40), size=n,
+                               replace=False)).astype(np.uint64) \
+            if n else np.empty(0, dtype=np.uint64)
+        if i % 3 == 0:
+            m = rng.integers(0, 2, size=n, dtype=np.uint8)
+        elif i % 3 == 1:
+            m = np.ones(n, dtype=np.uint8)   # keep all
+        else:
+            m = np.zeros(n, dtype=np.uint8)  # drop all
+        us_np.append(u)
+        masks_np.append(m)
+    us = [to_dev(u) for u in us_np]
+    masks = [torch.from_numpy(m.view(np.int8)).to("cuda:0").view(torch.uint8)
+             for m in masks_np]
+    if inplace:
+        outs, lens = eng.apply_filter_batch(us, masks)
+    else:
+        outs = [torch.empty_like(u) for u in us]
+        outs, lens = eng.apply_filter_batch(us, masks, outs)
+    for i, n in enumerate(sizes):
+        want = orc.apply_filter(us_np[i], masks_np[i].astype(bool))
+        assert lens[i] == want.size
+        assert to_np(outs[i][:lens[i]]).tolist() == want.tolist()
