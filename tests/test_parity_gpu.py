@@ -620,3 +620,35 @@ def test_apply_filter_batch_dev(eng, inplace):
         want = orc.apply_filter(us_np[i], masks_np[i].astype(bool))
         assert lens[i] == want.size
         assert to_np(outs[i][:lens[i]]).tolist() == want.tolist()
+
+
+# ---------- cfg 3 extreme Zipf shapes (BASELINE cfg 3 clamp [1k, 10M]) ----------
+
+def test_zipf_extreme_pair_shapes(eng):
+    """Tile-count skew inside one grid: 10M-element lists (≈9766 tiles)
+    batched next to 1k ones (1 tile), all three ops vs the oracle."""
+    rng = np.random.default_rng(SEED + 33)
+    def gen(n):
+        deltas = rng.integers(1, 17, size=n, dtype=np.uint64)
+        return np.cumsum(deltas).astype(np.uint64)
+    TEN_M = 10_000_000
+    shapes = [(TEN_M, TEN_M), (TEN_M, 1000), (1000, TEN_M), (1000, 1000),
+              (TEN_M, 100_000)]
+    us_np = [gen(n) for n, _ in shapes]
+    vs_np = [gen(m) for _, m in shapes]
+    us = [to_dev(u) for u in us_np]
+    vs = [to_dev(v) for v in vs_np]
+    for op_name in ["intersect", "merge", "difference"]:
+        if op_name == "intersect":
+            outs, lens = eng.intersect_pairs(us, vs)
+            want = [orc.intersect_with(u, v) for u, v in zip(us_np, vs_np)]
+        elif op_name == "merge":
+            outs, lens = eng.merge_pairs(us, vs)
+            want = [orc.merge_sorted([u, v]) for u, v in zip(us_np, vs_np)]
+        else:
+            outs, lens = eng.difference_pairs(us, vs)
+            want = [orc.difference(u, v) for u, v in zip(us_np, vs_np)]
+        for i in range(len(shapes)):
+            assert lens[i] == want[i].size, (op_name, i)
+            got = to_np(outs[i][:lens[i]])
+            assert np.array_equal(got, want[i]), (op_name, i)
