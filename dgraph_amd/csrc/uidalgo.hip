@@ -1864,10 +1864,19 @@ static int ws_reserve(ua_ctx *c, int slot, size_t bytes) {
     return UA_OK;
 }
 
+/* Which ops run the single-pass decoupled-lookback pipeline.
+ * MEASURED (r02, cfg2 192x1Mx1M): lookback doubles the intersect kernel
+ * (0.76 -> 1.47 ms) — a tile cannot write until every same-pair predecessor
+ * has finished its walk, so each tile's lifetime spans ~2 walk phases and
+ * 8 WGs/CU cannot hide the parked resolve.  For intersect/difference the
+ * staged scan+compact pipeline (outputs ~1% of inputs -> staging is free)
+ * is strictly better; for UNION the lookback removes a FULL second data
+ * pass (count+write -> one walk), which outweighs the same wait.  */
 #ifndef UA_LOOKBACK
-#define UA_LOOKBACK 1 /* 1 = single-pass decoupled-lookback pipeline (one tile
-                       * kernel per batch run, no scan/compact launches, union
-                       * walks once); 0 = the r01 staged scan+compact path */
+#define UA_LOOKBACK 0 /* intersect/diff: 1 = lookback, 0 = staged (default) */
+#endif
+#ifndef UA_LOOKBACK_UNION
+#define UA_LOOKBACK_UNION 1 /* union/merge-tree: single-pass lookback */
 #endif
 
 /* Acquire the ctx-level lookback flag array for T tiles with a fresh
@@ -2011,14 +2020,15 @@ static int run_batch_locked(ua_ctx *c, const ua_dpair *pairs, int n_pairs,
     if ((rc = ws_reserve(c, WS_TOFF, (total_tiles + 1) * sizeof(u64)))) return rc;
     if ((rc = ws_reserve(c, WS_POUT, (size_t)n_pairs * sizeof(u64)))) return rc;
 
+    const bool lb = (op == OP_UNION) ? (bool)UA_LOOKBACK_UNION : (bool)UA_LOOKBACK;
     u64 stage_stride = 0;
-#if !UA_LOOKBACK
-    if (op == OP_INTERSECT) stage_stride = UA_TILE / 2;
-    else if (op == OP_DIFF) stage_stride = UA_TILE;
-    if (stage_stride) {
-        if ((rc = ws_reserve(c, WS_STAGE, total_tiles * stage_stride * sizeof(u64)))) return rc;
+    if (!lb) {
+        if (op == OP_INTERSECT) stage_stride = UA_TILE / 2;
+        else if (op == OP_DIFF) stage_stride = UA_TILE;
+        if (stage_stride) {
+            if ((rc = ws_reserve(c, WS_STAGE, total_tiles * stage_stride * sizeof(u64)))) return rc;
+        }
     }
-#endif
 
     UaDesc *d_descs = (UaDesc *)c->ws[WS_DESC];
     u64 *d_tb = (u64 *)((u8 *)c->ws[WS_DESC] + descs_bytes);
@@ -2037,7 +2047,13 @@ static int run_batch_locked(ua_ctx *c, const ua_dpair *pairs, int n_pairs,
     /* zero sentinel so offs[total_tiles] = total output */
     HIP_TRY(hipMemsetAsync(d_tcnt + total_tiles, 0, sizeof(u32), c->stream));
 
-    if (total_tiles > 0) {
+    if (total_tiles == 0) {
+        /* every pair is empty (n+m == 0) */
+        HIP_TRY(hipStreamSynchronize(c->stream)); /* hostbuf is scoped */
+        for (int p = 0; p < n_pairs; p++) out_lens[p] = 0;
+        return UA_OK;
+    }
+    {
         u64 pblk = (total_tiles + UA_BLOCK - 1) / UA_BLOCK;
         hipLaunchKernelGGL(k_partition, dim3((u32)pblk), dim3(UA_BLOCK), 0, c->stream,
                            d_descs, d_tb, n_pairs, total_tiles, d_tpair, d_ta0, 0);
@@ -2063,96 +2079,85 @@ static int run_batch_locked(ua_ctx *c, const ua_dpair *pairs, int n_pairs,
             return UA_OK;
         }
 
-#if UA_LOOKBACK
-        u64 *d_lbf;
-        u64 gen;
-        if ((rc = lb_acquire_ws(c, total_tiles, &d_lbf, &gen))) return rc;
-        HIP_TRY(hipEventRecord(c->ev[0], c->stream));
-        if (op == OP_INTERSECT) {
-            launch_tiles<OP_INTERSECT, MODE_LOOKBACK>(c, d_descs, d_tpair, d_ta0,
+        bool two_kernels = false;
+        if (lb) {
+            u64 *d_lbf;
+            u64 gen;
+            if ((rc = lb_acquire_ws(c, total_tiles, &d_lbf, &gen))) return rc;
+            HIP_TRY(hipEventRecord(c->ev[0], c->stream));
+            if (op == OP_INTERSECT) {
+                launch_tiles<OP_INTERSECT, MODE_LOOKBACK>(c, d_descs, d_tpair, d_ta0,
+                                                          total_tiles, d_lbf, gen,
+                                                          nullptr, d_pout, nullptr);
+            } else if (op == OP_DIFF) {
+                launch_tiles<OP_DIFF, MODE_LOOKBACK>(c, d_descs, d_tpair, d_ta0,
+                                                     total_tiles, d_lbf, gen, nullptr,
+                                                     d_pout, nullptr);
+            } else {
+                launch_tiles<OP_UNION, MODE_LOOKBACK>(c, d_descs, d_tpair, d_ta0,
                                                       total_tiles, d_lbf, gen, nullptr,
                                                       d_pout, nullptr);
-        } else if (op == OP_DIFF) {
-            launch_tiles<OP_DIFF, MODE_LOOKBACK>(c, d_descs, d_tpair, d_ta0,
-                                                 total_tiles, d_lbf, gen, nullptr,
-                                                 d_pout, nullptr);
+            }
+            HIP_TRY(hipEventRecord(c->ev[1], c->stream));
         } else {
-            launch_tiles<OP_UNION, MODE_LOOKBACK>(c, d_descs, d_tpair, d_ta0,
-                                                  total_tiles, d_lbf, gen, nullptr,
-                                                  d_pout, nullptr);
+            HIP_TRY(hipEventRecord(c->ev[0], c->stream));
+            if (op == OP_INTERSECT) {
+                launch_tiles<OP_INTERSECT, MODE_STAGE>(c, d_descs, d_tpair, d_ta0,
+                                                       total_tiles, d_stage,
+                                                       stage_stride, d_tcnt, nullptr,
+                                                       nullptr);
+            } else if (op == OP_DIFF) {
+                launch_tiles<OP_DIFF, MODE_STAGE>(c, d_descs, d_tpair, d_ta0,
+                                                  total_tiles, d_stage, stage_stride,
+                                                  d_tcnt, nullptr, nullptr);
+            } else {
+                launch_tiles<OP_UNION, MODE_COUNT>(c, d_descs, d_tpair, d_ta0,
+                                                   total_tiles, nullptr, 0, d_tcnt,
+                                                   nullptr, nullptr);
+            }
+            HIP_TRY(hipEventRecord(c->ev[1], c->stream));
+
+            if ((rc = run_scan(c, d_tcnt, total_tiles + 1, d_toff))) return rc;
+            u64 *d_part = (u64 *)c->ws[WS_PARTIAL];
+
+            if (op == OP_UNION) {
+                two_kernels = true;
+                HIP_TRY(hipEventRecord(c->ev[2], c->stream));
+                launch_tiles<OP_UNION, MODE_WRITE>(c, d_descs, d_tpair, d_ta0,
+                                                   total_tiles, nullptr, 0, d_tcnt,
+                                                   d_toff, d_part);
+                HIP_TRY(hipEventRecord(c->ev[3], c->stream));
+            } else {
+                u64 cblk = (total_tiles + 15) / 16;
+                hipLaunchKernelGGL(k_compact, dim3((u32)cblk), dim3(UA_BLOCK), 0,
+                                   c->stream, d_descs, d_tpair, d_tcnt, d_toff, d_part,
+                                   d_stage, stage_stride, total_tiles, op);
+            }
+            u64 poutblk = ((u64)n_pairs + UA_BLOCK - 1) / UA_BLOCK;
+            hipLaunchKernelGGL(k_pair_out, dim3((u32)poutblk), dim3(UA_BLOCK), 0,
+                               c->stream, d_toff, d_part, d_tb, n_pairs, d_pout);
         }
-        HIP_TRY(hipEventRecord(c->ev[1], c->stream));
-    }
-    HIP_TRY(hipMemcpyAsync(out_lens, d_pout, (size_t)n_pairs * sizeof(u64),
-                           hipMemcpyDeviceToHost, c->stream));
-    HIP_TRY(hipStreamSynchronize(c->stream));
-    HIP_TRY(hipGetLastError());
-    /* pairs with zero tiles (n+m == 0) never publish a length */
-    for (int p = 0; p < n_pairs; p++)
-        if (pairs[p].n + pairs[p].m == 0) out_lens[p] = 0;
-    (void)d_tcnt;
-    (void)d_toff;
-    (void)d_stage;
-    (void)stage_stride;
-#else
-        HIP_TRY(hipEventRecord(c->ev[0], c->stream));
-        if (op == OP_INTERSECT) {
-            launch_tiles<OP_INTERSECT, MODE_STAGE>(c, d_descs, d_tpair, d_ta0,
-                                                   total_tiles, d_stage, stage_stride,
-                                                   d_tcnt, nullptr, nullptr);
-        } else if (op == OP_DIFF) {
-            launch_tiles<OP_DIFF, MODE_STAGE>(c, d_descs, d_tpair, d_ta0, total_tiles,
-                                              d_stage, stage_stride, d_tcnt, nullptr,
-                                              nullptr);
-        } else {
-            launch_tiles<OP_UNION, MODE_COUNT>(c, d_descs, d_tpair, d_ta0, total_tiles,
-                                               nullptr, 0, d_tcnt, nullptr, nullptr);
+        HIP_TRY(hipMemcpyAsync(out_lens, d_pout, (size_t)n_pairs * sizeof(u64),
+                               hipMemcpyDeviceToHost, c->stream));
+        HIP_TRY(hipStreamSynchronize(c->stream));
+        HIP_TRY(hipGetLastError());
+        if (lb) {
+            /* pairs with zero tiles (n+m == 0) never publish a length */
+            for (int p = 0; p < n_pairs; p++)
+                if (pairs[p].n + pairs[p].m == 0) out_lens[p] = 0;
         }
-        HIP_TRY(hipEventRecord(c->ev[1], c->stream));
 
-        if ((rc = run_scan(c, d_tcnt, total_tiles + 1, d_toff))) return rc;
-        u64 *d_part = (u64 *)c->ws[WS_PARTIAL];
-
-        if (op == OP_UNION) {
-            HIP_TRY(hipEventRecord(c->ev[2], c->stream));
-            launch_tiles<OP_UNION, MODE_WRITE>(c, d_descs, d_tpair, d_ta0, total_tiles,
-                                               nullptr, 0, d_tcnt, d_toff, d_part);
-            HIP_TRY(hipEventRecord(c->ev[3], c->stream));
-        } else {
-            u64 cblk = (total_tiles + 15) / 16;
-            hipLaunchKernelGGL(k_compact, dim3((u32)cblk), dim3(UA_BLOCK), 0, c->stream,
-                               d_descs, d_tpair, d_tcnt, d_toff, d_part, d_stage,
-                               stage_stride, total_tiles, op);
-        }
-    } else {
-        if ((rc = ws_reserve(c, WS_PARTIAL, sizeof(u64)))) return rc;
-        HIP_TRY(hipMemsetAsync(d_toff, 0, (total_tiles + 1) * sizeof(u64), c->stream));
-        HIP_TRY(hipMemsetAsync(c->ws[WS_PARTIAL], 0, sizeof(u64), c->stream));
-    }
-
-    u64 poutblk = ((u64)n_pairs + UA_BLOCK - 1) / UA_BLOCK;
-    hipLaunchKernelGGL(k_pair_out, dim3((u32)poutblk), dim3(UA_BLOCK), 0, c->stream,
-                       d_toff, (const u64 *)c->ws[WS_PARTIAL], d_tb, n_pairs, d_pout);
-    HIP_TRY(hipMemcpyAsync(out_lens, d_pout, (size_t)n_pairs * sizeof(u64),
-                           hipMemcpyDeviceToHost, c->stream));
-    HIP_TRY(hipStreamSynchronize(c->stream));
-    HIP_TRY(hipGetLastError());
-#endif
-
-    /* stats: HIP-event time of the dominant (tile) kernel(s) on this stream */
-    if (total_tiles > 0) {
+        /* stats: HIP-event time of the dominant (tile) kernel(s) */
         float ms = 0.f;
         HIP_TRY(hipEventElapsedTime(&ms, c->ev[0], c->ev[1]));
         c->kernel_ms += ms;
         c->n_launches += 1;
-#if !UA_LOOKBACK
-        if (op == OP_UNION) {
+        if (two_kernels) {
             float ms2 = 0.f;
             HIP_TRY(hipEventElapsedTime(&ms2, c->ev[2], c->ev[3]));
             c->kernel_ms += ms2;
             c->n_launches += 1;
         }
-#endif
     }
     u64 out_elems = 0;
     for (int p = 0; p < n_pairs; p++) out_elems += out_lens[p];
@@ -2189,6 +2194,10 @@ struct ua_batch {
     u64 *d_pout = nullptr;
     u64 *d_lbf = nullptr;    /* lookback flag array [total_tiles] */
     u32 lb_gen = 0;
+    /* hipGraph capture of the staged pipeline (per op); out_lens land in the
+     * pinned h_pout so the captured D2H copy has a fixed destination */
+    hipGraphExec_t gexec[3] = {};
+    u64 *h_pout = nullptr;
 };
 
 static size_t align16(size_t x) { return (x + 15) & ~(size_t)15; }
@@ -2274,9 +2283,59 @@ extern "C" int ua_batch_create(ua_ctx *c, const ua_dpair *pairs, int n_pairs,
 extern "C" void ua_batch_destroy(ua_ctx *c, ua_batch *b) {
     if (!b) return;
     (void)hipSetDevice(c->device);
+    for (int i = 0; i < 3; i++)
+        if (b->gexec[i]) (void)hipGraphExecDestroy(b->gexec[i]);
+    if (b->h_pout) (void)hipHostFree(b->h_pout);
     if (b->mem) (void)hipFree(b->mem);
     if (b->d_stage) (void)hipFree(b->d_stage);
     delete b;
+}
+
+#ifndef UA_GRAPH
+#define UA_GRAPH 1 /* hipGraph-capture the staged prepared-batch pipeline
+                    * (5 launches -> one replay); 0 = eager launches */
+#endif
+
+/* the staged pipeline sequence on b (enqueue only — also what gets graph-
+ * captured); lens land in b->d_pout and (when h_pout) the pinned copy */
+static int batch_staged_seq(ua_ctx *c, ua_batch *b, int kop, u64 stride,
+                            u64 *host_pout) {
+    u64 T = b->total_tiles;
+    HIP_TRY(hipEventRecord(c->ev[0], c->stream));
+    if (kop == OP_INTERSECT) {
+        launch_tiles<OP_INTERSECT, MODE_STAGE>(c, b->d_descs, b->d_tpair, b->d_ta0,
+                                               T, b->d_stage, stride, b->d_tcnt,
+                                               nullptr, nullptr);
+    } else if (kop == OP_DIFF) {
+        launch_tiles<OP_DIFF, MODE_STAGE>(c, b->d_descs, b->d_tpair, b->d_ta0, T,
+                                          b->d_stage, stride, b->d_tcnt, nullptr,
+                                          nullptr);
+    } else {
+        launch_tiles<OP_UNION, MODE_COUNT>(c, b->d_descs, b->d_tpair, b->d_ta0, T,
+                                           nullptr, 0, b->d_tcnt, nullptr, nullptr);
+    }
+    HIP_TRY(hipEventRecord(c->ev[1], c->stream));
+    hipLaunchKernelGGL(k_scan1, dim3((u32)b->nchunks), dim3(UA_BLOCK), 0, c->stream,
+                       b->d_tcnt, T + 1, b->d_toff, b->d_part);
+    hipLaunchKernelGGL(k_scan2, dim3(1), dim3(UA_BLOCK), 0, c->stream, b->d_part,
+                       b->nchunks);
+    if (kop == OP_UNION) {
+        HIP_TRY(hipEventRecord(c->ev[2], c->stream));
+        launch_tiles<OP_UNION, MODE_WRITE>(c, b->d_descs, b->d_tpair, b->d_ta0, T,
+                                           nullptr, 0, b->d_tcnt, b->d_toff,
+                                           b->d_part);
+        HIP_TRY(hipEventRecord(c->ev[3], c->stream));
+    } else {
+        hipLaunchKernelGGL(k_compact, dim3((u32)((T + 15) / 16)), dim3(UA_BLOCK), 0,
+                           c->stream, b->d_descs, b->d_tpair, b->d_tcnt, b->d_toff,
+                           b->d_part, b->d_stage, stride, T, kop);
+    }
+    u64 poutblk = ((u64)b->n_pairs + UA_BLOCK - 1) / UA_BLOCK;
+    hipLaunchKernelGGL(k_pair_out, dim3((u32)poutblk), dim3(UA_BLOCK), 0, c->stream,
+                       b->d_toff, b->d_part, b->d_tb, b->n_pairs, b->d_pout);
+    HIP_TRY(hipMemcpyAsync(host_pout, b->d_pout, (size_t)b->n_pairs * sizeof(u64),
+                           hipMemcpyDeviceToHost, c->stream));
+    return UA_OK;
 }
 
 extern "C" int ua_batch_run(ua_ctx *c, ua_batch *b, int op, uint64_t *out_lens) {
@@ -2289,8 +2348,13 @@ extern "C" int ua_batch_run(ua_ctx *c, ua_batch *b, int op, uint64_t *out_lens) 
                                          : -1;
     if (kop < 0) return UA_ERR_INVALID;
     u64 T = b->total_tiles;
-#if UA_LOOKBACK
-    if (T > 0) {
+    if (T == 0) {
+        for (int p = 0; p < b->n_pairs; p++) out_lens[p] = 0;
+        return UA_OK;
+    }
+    const bool lb = (kop == OP_UNION) ? (bool)UA_LOOKBACK_UNION : (bool)UA_LOOKBACK;
+    bool two_kernels = false;
+    if (lb) {
         if (b->lb_gen >= UA_LB_GEN_MAX) {
             HIP_TRY(hipMemsetAsync(b->d_lbf, 0, (T + 1) * sizeof(u64), c->stream));
             b->lb_gen = 0;
@@ -2312,70 +2376,76 @@ extern "C" int ua_batch_run(ua_ctx *c, ua_batch *b, int op, uint64_t *out_lens) 
                                                   nullptr);
         }
         HIP_TRY(hipEventRecord(c->ev[1], c->stream));
-    }
+        HIP_TRY(hipMemcpyAsync(out_lens, b->d_pout, (size_t)b->n_pairs * sizeof(u64),
+                               hipMemcpyDeviceToHost, c->stream));
+        HIP_TRY(hipStreamSynchronize(c->stream));
+        HIP_TRY(hipGetLastError());
+        /* zero-tile pairs stay 0 from the create-time d_pout memset */
+    } else {
+        two_kernels = (kop == OP_UNION);
+        u64 stride = (kop == OP_INTERSECT) ? UA_TILE / 2 : UA_TILE;
+        if (kop != OP_UNION && !b->d_stage) {
+            hipError_t e = hipMalloc((void **)&b->d_stage,
+                                     (T ? T : 1) * UA_TILE * sizeof(u64));
+            if (e != hipSuccess) {
+                g_last_hip = e;
+                return UA_ERR_NOMEM;
+            }
+        }
+        if (!b->h_pout) {
+            hipError_t e = hipHostMalloc((void **)&b->h_pout,
+                                         (size_t)b->n_pairs * sizeof(u64));
+            if (e != hipSuccess) {
+                g_last_hip = e;
+                return UA_ERR_NOMEM;
+            }
+        }
+        int rc;
+#if UA_GRAPH
+        if (!b->gexec[kop]) {
+            /* capture once; a capture failure (e.g. event-record nodes
+             * unsupported) falls back to eager for this run */
+            hipError_t ce = hipStreamBeginCapture(c->stream,
+                                                  hipStreamCaptureModeThreadLocal);
+            if (ce == hipSuccess) {
+                rc = batch_staged_seq(c, b, kop, stride, b->h_pout);
+                hipGraph_t gr = nullptr;
+                hipError_t ee = hipStreamEndCapture(c->stream, &gr);
+                if (rc == UA_OK && ee == hipSuccess && gr) {
+                    hipError_t ie = hipGraphInstantiate(&b->gexec[kop], gr, nullptr,
+                                                        nullptr, 0);
+                    (void)hipGraphDestroy(gr);
+                    if (ie != hipSuccess) b->gexec[kop] = nullptr;
+                } else {
+                    if (gr) (void)hipGraphDestroy(gr);
+                    if (rc != UA_OK) return rc;
+                }
+            }
+        }
+        if (b->gexec[kop]) {
+            HIP_TRY(hipGraphLaunch(b->gexec[kop], c->stream));
+        } else {
+            if ((rc = batch_staged_seq(c, b, kop, stride, b->h_pout))) return rc;
+        }
 #else
-    u64 stride = (kop == OP_INTERSECT) ? UA_TILE / 2 : UA_TILE;
-    if (kop != OP_UNION && !b->d_stage) {
-        hipError_t e = hipMalloc((void **)&b->d_stage, (T ? T : 1) * UA_TILE * sizeof(u64));
-        if (e != hipSuccess) {
-            g_last_hip = e;
-            return UA_ERR_NOMEM;
-        }
-    }
-    if (T > 0) {
-        HIP_TRY(hipEventRecord(c->ev[0], c->stream));
-        if (kop == OP_INTERSECT) {
-            launch_tiles<OP_INTERSECT, MODE_STAGE>(c, b->d_descs, b->d_tpair, b->d_ta0,
-                                                   T, b->d_stage, stride, b->d_tcnt,
-                                                   nullptr, nullptr);
-        } else if (kop == OP_DIFF) {
-            launch_tiles<OP_DIFF, MODE_STAGE>(c, b->d_descs, b->d_tpair, b->d_ta0, T,
-                                              b->d_stage, stride, b->d_tcnt, nullptr,
-                                              nullptr);
-        } else {
-            launch_tiles<OP_UNION, MODE_COUNT>(c, b->d_descs, b->d_tpair, b->d_ta0, T,
-                                               nullptr, 0, b->d_tcnt, nullptr, nullptr);
-        }
-        HIP_TRY(hipEventRecord(c->ev[1], c->stream));
-
-        hipLaunchKernelGGL(k_scan1, dim3((u32)b->nchunks), dim3(UA_BLOCK), 0, c->stream,
-                           b->d_tcnt, T + 1, b->d_toff, b->d_part);
-        hipLaunchKernelGGL(k_scan2, dim3(1), dim3(UA_BLOCK), 0, c->stream, b->d_part,
-                           b->nchunks);
-        if (kop == OP_UNION) {
-            HIP_TRY(hipEventRecord(c->ev[2], c->stream));
-            launch_tiles<OP_UNION, MODE_WRITE>(c, b->d_descs, b->d_tpair, b->d_ta0, T,
-                                               nullptr, 0, b->d_tcnt, b->d_toff,
-                                               b->d_part);
-            HIP_TRY(hipEventRecord(c->ev[3], c->stream));
-        } else {
-            hipLaunchKernelGGL(k_compact, dim3((u32)((T + 15) / 16)), dim3(UA_BLOCK), 0,
-                               c->stream, b->d_descs, b->d_tpair, b->d_tcnt, b->d_toff,
-                               b->d_part, b->d_stage, stride, T, kop);
-        }
-    }
-    u64 poutblk = ((u64)b->n_pairs + UA_BLOCK - 1) / UA_BLOCK;
-    hipLaunchKernelGGL(k_pair_out, dim3((u32)poutblk), dim3(UA_BLOCK), 0, c->stream,
-                       b->d_toff, b->d_part, b->d_tb, b->n_pairs, b->d_pout);
+        if ((rc = batch_staged_seq(c, b, kop, stride, b->h_pout))) return rc;
 #endif
-    HIP_TRY(hipMemcpyAsync(out_lens, b->d_pout, (size_t)b->n_pairs * sizeof(u64),
-                           hipMemcpyDeviceToHost, c->stream));
-    HIP_TRY(hipStreamSynchronize(c->stream));
-    HIP_TRY(hipGetLastError());
+        HIP_TRY(hipStreamSynchronize(c->stream));
+        HIP_TRY(hipGetLastError());
+        memcpy(out_lens, b->h_pout, (size_t)b->n_pairs * sizeof(u64));
+    }
 
-    if (T > 0) {
-        float ms = 0.f;
-        HIP_TRY(hipEventElapsedTime(&ms, c->ev[0], c->ev[1]));
+    float ms = 0.f;
+    if (hipEventElapsedTime(&ms, c->ev[0], c->ev[1]) == hipSuccess) {
         c->kernel_ms += ms;
         c->n_launches += 1;
-#if !UA_LOOKBACK
-        if (kop == OP_UNION) {
-            float ms2 = 0.f;
-            HIP_TRY(hipEventElapsedTime(&ms2, c->ev[2], c->ev[3]));
+    }
+    if (two_kernels) {
+        float ms2 = 0.f;
+        if (hipEventElapsedTime(&ms2, c->ev[2], c->ev[3]) == hipSuccess) {
             c->kernel_ms += ms2;
             c->n_launches += 1;
         }
-#endif
     }
     u64 out_elems = 0;
     for (int p = 0; p < b->n_pairs; p++) out_elems += out_lens[p];
@@ -2719,7 +2789,7 @@ extern "C" int ua_merge_k_dev(ua_ctx *c, const uint64_t *const *lists,
     if ((rc = ws_reserve(c, WS_TOFF, (max_tiles + 1) * sizeof(u64)))) return rc;
     u64 max_chunks = (max_tiles + 1 + UA_SCAN_CHUNK - 1) / UA_SCAN_CHUNK;
     if ((rc = ws_reserve(c, WS_PARTIAL, (max_chunks + 1) * sizeof(u64)))) return rc;
-#if UA_LOOKBACK
+#if UA_LOOKBACK_UNION
     /* pre-size the lookback flag array so per-round acquires never realloc
      * mid-enqueue (hipFree would drain the stream) */
     if ((rc = ws_reserve(c, WS_LBF, (max_tiles + 1) * sizeof(u64)))) return rc;
@@ -2743,7 +2813,7 @@ extern "C" int ua_merge_k_dev(ua_ctx *c, const uint64_t *const *lists,
         u64 mk = ((u64)rd.npair + UA_BLOCK - 1) / UA_BLOCK;
         hipLaunchKernelGGL(k_make_descs, dim3((u32)mk), dim3(UA_BLOCK), 0, c->stream,
                            d_descs, d_lens[lcur], rd.nk_prev, rd.npair);
-#if UA_LOOKBACK
+#if UA_LOOKBACK_UNION
         /* zero-capacity pairs have no tiles and never publish a length */
         HIP_TRY(hipMemsetAsync(d_lens[lcur ^ 1], 0, (size_t)rd.npair * sizeof(u64),
                                c->stream));
@@ -2802,7 +2872,7 @@ extern "C" int ua_merge_k_dev(ua_ctx *c, const uint64_t *const *lists,
     HIP_TRY(hipStreamSynchronize(c->stream));
     HIP_TRY(hipEventElapsedTime(&ms, c->ev[0], c->ev[1]));
     c->kernel_ms += ms;
-#if UA_LOOKBACK
+#if UA_LOOKBACK_UNION
     c->n_launches += R;
     c->bytes_algo += 8 * 2 * cap_work; /* capacity upper bound (in + out, one pass) */
 #else
