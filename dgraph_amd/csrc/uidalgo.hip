@@ -29,6 +29,8 @@
 
 #include <algorithm>
 #include <cstdint>
+#include <cstdio>
+#include <cstdlib>
 #include <cstring>
 #include <mutex>
 #include <vector>
@@ -2197,6 +2199,7 @@ struct ua_batch {
     /* hipGraph capture of the staged pipeline (per op); out_lens land in the
      * pinned h_pout so the captured D2H copy has a fixed destination */
     hipGraphExec_t gexec[3] = {};
+    bool no_graph = false; /* capture/replay failed on this box: stay eager */
     u64 *h_pout = nullptr;
 };
 
@@ -2402,9 +2405,10 @@ extern "C" int ua_batch_run(ua_ctx *c, ua_batch *b, int op, uint64_t *out_lens) 
         }
         int rc;
 #if UA_GRAPH
-        if (!b->gexec[kop]) {
-            /* capture once; a capture failure (e.g. event-record nodes
-             * unsupported) falls back to eager for this run */
+        if (!b->gexec[kop] && !b->no_graph) {
+            /* capture once and VALIDATE with a launch+sync; any failure
+             * (e.g. event-record nodes unsupported on a ROCm build) makes
+             * this batch permanently eager — self-healing, never fatal */
             hipError_t ce = hipStreamBeginCapture(c->stream,
                                                   hipStreamCaptureModeThreadLocal);
             if (ce == hipSuccess) {
@@ -2414,17 +2418,47 @@ extern "C" int ua_batch_run(ua_ctx *c, ua_batch *b, int op, uint64_t *out_lens) 
                 if (rc == UA_OK && ee == hipSuccess && gr) {
                     hipError_t ie = hipGraphInstantiate(&b->gexec[kop], gr, nullptr,
                                                         nullptr, 0);
+                    if (ie == hipSuccess) {
+                        hipError_t le = hipGraphLaunch(b->gexec[kop], c->stream);
+                        if (le == hipSuccess) le = hipStreamSynchronize(c->stream);
+                        if (le != hipSuccess) {
+                            if (getenv("UA_DEBUG"))
+                                fprintf(stderr, "[ua] graph launch failed: %s\n",
+                                        hipGetErrorName(le));
+                            (void)hipGraphExecDestroy(b->gexec[kop]);
+                            b->gexec[kop] = nullptr;
+                            b->no_graph = true;
+                            (void)hipGetLastError();
+                        }
+                    } else {
+                        if (getenv("UA_DEBUG"))
+                            fprintf(stderr, "[ua] graph instantiate failed: %s\n",
+                                    hipGetErrorName(ie));
+                        b->gexec[kop] = nullptr;
+                        b->no_graph = true;
+                    }
                     (void)hipGraphDestroy(gr);
-                    if (ie != hipSuccess) b->gexec[kop] = nullptr;
                 } else {
                     if (gr) (void)hipGraphDestroy(gr);
+                    b->no_graph = true;
+                    (void)hipGetLastError();
                     if (rc != UA_OK) return rc;
                 }
+            } else {
+                b->no_graph = true;
+                (void)hipGetLastError();
             }
         }
         if (b->gexec[kop]) {
-            HIP_TRY(hipGraphLaunch(b->gexec[kop], c->stream));
-        } else {
+            hipError_t le = hipGraphLaunch(b->gexec[kop], c->stream);
+            if (le != hipSuccess) {
+                (void)hipGraphExecDestroy(b->gexec[kop]);
+                b->gexec[kop] = nullptr;
+                b->no_graph = true;
+                (void)hipGetLastError();
+            }
+        }
+        if (!b->gexec[kop]) {
             if ((rc = batch_staged_seq(c, b, kop, stride, b->h_pout))) return rc;
         }
 #else
