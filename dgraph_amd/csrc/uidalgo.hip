@@ -1832,7 +1832,13 @@ extern "C" int ua_ctx_create(ua_ctx **out, int device) {
     c->device = device;
     HIP_TRY(hipSetDevice(device));
     HIP_TRY(hipStreamCreate(&c->stream));
-    for (int i = 0; i < 4; i++) HIP_TRY(hipEventCreate(&c->ev[i]));
+    for (int i = 0; i < 4; i++) {
+        HIP_TRY(hipEventCreate(&c->ev[i]));
+        /* record once eagerly: an event that has never been recorded cannot
+         * be captured into a graph on this ROCm (invalid resource handle) */
+        HIP_TRY(hipEventRecord(c->ev[i], c->stream));
+    }
+    HIP_TRY(hipStreamSynchronize(c->stream));
     *out = c;
     return UA_OK;
 }
@@ -2439,10 +2445,15 @@ extern "C" int ua_batch_run(ua_ctx *c, ua_batch *b, int op, uint64_t *out_lens) 
                     }
                     (void)hipGraphDestroy(gr);
                 } else {
+                    /* enqueue error mid-capture (e.g. event record rejected)
+                     * or capture teardown failure: go eager — the capture
+                     * recorded nothing, the stream is clean after EndCapture */
+                    if (getenv("UA_DEBUG"))
+                        fprintf(stderr, "[ua] capture failed (rc=%d, end=%s)\n", rc,
+                                hipGetErrorName(ee));
                     if (gr) (void)hipGraphDestroy(gr);
                     b->no_graph = true;
                     (void)hipGetLastError();
-                    if (rc != UA_OK) return rc;
                 }
             } else {
                 b->no_graph = true;
