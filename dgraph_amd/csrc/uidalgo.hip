@@ -1884,7 +1884,10 @@ static int ws_reserve(ua_ctx *c, int slot, size_t bytes) {
 #define UA_LOOKBACK 0 /* intersect/diff: 1 = lookback, 0 = staged (default) */
 #endif
 #ifndef UA_LOOKBACK_UNION
-#define UA_LOOKBACK_UNION 1 /* union/merge-tree: single-pass lookback */
+#define UA_LOOKBACK_UNION 0 /* union/merge-tree single-pass lookback: MEASURED
+                             * WORSE on cfg3 (125 vs 145 G elems/s) — the
+                             * same-pair predecessor wait costs more than the
+                             * saved second read on long tile chains */
 #endif
 
 /* Acquire the ctx-level lookback flag array for T tiles with a fresh
@@ -2301,14 +2304,16 @@ extern "C" void ua_batch_destroy(ua_ctx *c, ua_batch *b) {
 }
 
 #ifndef UA_GRAPH
-#define UA_GRAPH 1 /* hipGraph-capture the staged prepared-batch pipeline
-                    * (5 launches -> one replay); 0 = eager launches */
+#define UA_GRAPH 1 /* hipGraph-capture the staged pipeline's AUX TAIL
+                    * (scan/compact/pair_out/D2H -> one replay).  The tile
+                    * kernel stays an eager launch so its HIP-event timing
+                    * (bench.py roofline leg) remains live — event-record
+                    * nodes inside replayed graphs do not update events on
+                    * this ROCm.  0 = eager launches */
 #endif
 
-/* the staged pipeline sequence on b (enqueue only — also what gets graph-
- * captured); lens land in b->d_pout and (when h_pout) the pinned copy */
-static int batch_staged_seq(ua_ctx *c, ua_batch *b, int kop, u64 stride,
-                            u64 *host_pout) {
+/* the staged TILE kernel (eager, event-timed) */
+static int batch_tiles_seq(ua_ctx *c, ua_batch *b, int kop, u64 stride) {
     u64 T = b->total_tiles;
     HIP_TRY(hipEventRecord(c->ev[0], c->stream));
     if (kop == OP_INTERSECT) {
@@ -2324,6 +2329,15 @@ static int batch_staged_seq(ua_ctx *c, ua_batch *b, int kop, u64 stride,
                                            nullptr, 0, b->d_tcnt, nullptr, nullptr);
     }
     HIP_TRY(hipEventRecord(c->ev[1], c->stream));
+    return UA_OK;
+}
+
+/* the aux tail: scan + write/compact + pair_out + D2H of lens (captured
+ * into a hipGraph for intersect/diff; union's WRITE pass is a second tile
+ * kernel that needs live events, so union stays eager throughout) */
+static int batch_tail_seq(ua_ctx *c, ua_batch *b, int kop, u64 stride,
+                          u64 *host_pout) {
+    u64 T = b->total_tiles;
     hipLaunchKernelGGL(k_scan1, dim3((u32)b->nchunks), dim3(UA_BLOCK), 0, c->stream,
                        b->d_tcnt, T + 1, b->d_toff, b->d_part);
     hipLaunchKernelGGL(k_scan2, dim3(1), dim3(UA_BLOCK), 0, c->stream, b->d_part,
@@ -2410,71 +2424,75 @@ extern "C" int ua_batch_run(ua_ctx *c, ua_batch *b, int op, uint64_t *out_lens) 
             }
         }
         int rc;
+        if ((rc = batch_tiles_seq(c, b, kop, stride))) return rc;
+        bool ran_tail = false;
 #if UA_GRAPH
-        if (!b->gexec[kop] && !b->no_graph) {
-            /* capture once and VALIDATE with a launch+sync; any failure
-             * (e.g. event-record nodes unsupported on a ROCm build) makes
-             * this batch permanently eager — self-healing, never fatal */
-            hipError_t ce = hipStreamBeginCapture(c->stream,
-                                                  hipStreamCaptureModeThreadLocal);
-            if (ce == hipSuccess) {
-                rc = batch_staged_seq(c, b, kop, stride, b->h_pout);
-                hipGraph_t gr = nullptr;
-                hipError_t ee = hipStreamEndCapture(c->stream, &gr);
-                if (rc == UA_OK && ee == hipSuccess && gr) {
-                    hipError_t ie = hipGraphInstantiate(&b->gexec[kop], gr, nullptr,
-                                                        nullptr, 0);
-                    if (ie == hipSuccess) {
-                        hipError_t le = hipGraphLaunch(b->gexec[kop], c->stream);
-                        if (le == hipSuccess) le = hipStreamSynchronize(c->stream);
-                        if (le != hipSuccess) {
+        if (kop != OP_UNION) { /* union's tail holds an event-timed kernel */
+            if (!b->gexec[kop] && !b->no_graph) {
+                /* capture the tail once and VALIDATE with a launch+sync; any
+                 * failure makes this batch permanently eager — self-healing,
+                 * never fatal.  The one validation launch IS this run's tail
+                 * (capture itself executes nothing). */
+                hipError_t ce = hipStreamBeginCapture(
+                    c->stream, hipStreamCaptureModeThreadLocal);
+                if (ce == hipSuccess) {
+                    rc = batch_tail_seq(c, b, kop, stride, b->h_pout);
+                    hipGraph_t gr = nullptr;
+                    hipError_t ee = hipStreamEndCapture(c->stream, &gr);
+                    if (rc == UA_OK && ee == hipSuccess && gr) {
+                        hipError_t ie = hipGraphInstantiate(&b->gexec[kop], gr,
+                                                            nullptr, nullptr, 0);
+                        if (ie == hipSuccess) {
+                            hipError_t le = hipGraphLaunch(b->gexec[kop], c->stream);
+                            if (le == hipSuccess)
+                                le = hipStreamSynchronize(c->stream);
+                            if (le == hipSuccess) {
+                                ran_tail = true;
+                            } else {
+                                if (getenv("UA_DEBUG"))
+                                    fprintf(stderr, "[ua] graph launch failed: %s\n",
+                                            hipGetErrorName(le));
+                                (void)hipGraphExecDestroy(b->gexec[kop]);
+                                b->gexec[kop] = nullptr;
+                                b->no_graph = true;
+                                (void)hipGetLastError();
+                            }
+                        } else {
                             if (getenv("UA_DEBUG"))
-                                fprintf(stderr, "[ua] graph launch failed: %s\n",
-                                        hipGetErrorName(le));
-                            (void)hipGraphExecDestroy(b->gexec[kop]);
+                                fprintf(stderr, "[ua] graph instantiate failed: %s\n",
+                                        hipGetErrorName(ie));
                             b->gexec[kop] = nullptr;
                             b->no_graph = true;
-                            (void)hipGetLastError();
                         }
+                        (void)hipGraphDestroy(gr);
                     } else {
                         if (getenv("UA_DEBUG"))
-                            fprintf(stderr, "[ua] graph instantiate failed: %s\n",
-                                    hipGetErrorName(ie));
-                        b->gexec[kop] = nullptr;
+                            fprintf(stderr, "[ua] capture failed (rc=%d, end=%s)\n",
+                                    rc, hipGetErrorName(ee));
+                        if (gr) (void)hipGraphDestroy(gr);
                         b->no_graph = true;
+                        (void)hipGetLastError();
                     }
-                    (void)hipGraphDestroy(gr);
                 } else {
-                    /* enqueue error mid-capture (e.g. event record rejected)
-                     * or capture teardown failure: go eager — the capture
-                     * recorded nothing, the stream is clean after EndCapture */
-                    if (getenv("UA_DEBUG"))
-                        fprintf(stderr, "[ua] capture failed (rc=%d, end=%s)\n", rc,
-                                hipGetErrorName(ee));
-                    if (gr) (void)hipGraphDestroy(gr);
                     b->no_graph = true;
                     (void)hipGetLastError();
                 }
-            } else {
-                b->no_graph = true;
-                (void)hipGetLastError();
+            } else if (b->gexec[kop]) {
+                hipError_t le = hipGraphLaunch(b->gexec[kop], c->stream);
+                if (le == hipSuccess) {
+                    ran_tail = true;
+                } else {
+                    (void)hipGraphExecDestroy(b->gexec[kop]);
+                    b->gexec[kop] = nullptr;
+                    b->no_graph = true;
+                    (void)hipGetLastError();
+                }
             }
         }
-        if (b->gexec[kop]) {
-            hipError_t le = hipGraphLaunch(b->gexec[kop], c->stream);
-            if (le != hipSuccess) {
-                (void)hipGraphExecDestroy(b->gexec[kop]);
-                b->gexec[kop] = nullptr;
-                b->no_graph = true;
-                (void)hipGetLastError();
-            }
-        }
-        if (!b->gexec[kop]) {
-            if ((rc = batch_staged_seq(c, b, kop, stride, b->h_pout))) return rc;
-        }
-#else
-        if ((rc = batch_staged_seq(c, b, kop, stride, b->h_pout))) return rc;
 #endif
+        if (!ran_tail) {
+            if ((rc = batch_tail_seq(c, b, kop, stride, b->h_pout))) return rc;
+        }
         HIP_TRY(hipStreamSynchronize(c->stream));
         HIP_TRY(hipGetLastError());
         memcpy(out_lens, b->h_pout, (size_t)b->n_pairs * sizeof(u64));
