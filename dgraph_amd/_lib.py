@@ -112,6 +112,7 @@ def lib():
         L.ua_intersect_batch_dev.argtypes = [C.c_void_p, C.POINTER(UaDPair), C.c_int, _u64p]
         L.ua_batch_create.argtypes = [C.c_void_p, C.POINTER(UaDPair), C.c_int, _voidpp]
         L.ua_batch_run.argtypes = [C.c_void_p, C.c_void_p, C.c_int, _u64p]
+        L.ua_batch_run_n.argtypes = [C.c_void_p, C.c_void_p, C.c_int, C.c_int, _u64p]
         L.ua_batch_destroy.argtypes = [C.c_void_p, C.c_void_p]
         L.ua_batch_destroy.restype = None
         L.ua_merge_batch_dev.argtypes = [C.c_void_p, C.POINTER(UaDPair), C.c_int, _u64p]

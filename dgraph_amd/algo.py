@@ -412,6 +412,13 @@ class Batch:
         check(lib().ua_batch_run(self._eng._ctx, self._h, op, lens))
         return [int(lens[i]) for i in range(self.n_pairs)]
 
+    def run_n(self, op, n_runs):
+        """n_runs passes enqueued back-to-back, ONE sync at the end (the
+        repeated-query serving shape — no host round-trip between runs)."""
+        lens = (C.c_uint64 * self.n_pairs)()
+        check(lib().ua_batch_run_n(self._eng._ctx, self._h, op, n_runs, lens))
+        return [int(lens[i]) for i in range(self.n_pairs)]
+
     def close(self):
         if self._h:
             lib().ua_batch_destroy(self._eng._ctx, self._h)

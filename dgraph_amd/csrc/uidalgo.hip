@@ -2313,9 +2313,10 @@ extern "C" void ua_batch_destroy(ua_ctx *c, ua_batch *b) {
 #endif
 
 /* the staged TILE kernel (eager, event-timed) */
-static int batch_tiles_seq(ua_ctx *c, ua_batch *b, int kop, u64 stride) {
+static int batch_tiles_seq(ua_ctx *c, ua_batch *b, int kop, u64 stride,
+                           bool record_events) {
     u64 T = b->total_tiles;
-    HIP_TRY(hipEventRecord(c->ev[0], c->stream));
+    if (record_events) HIP_TRY(hipEventRecord(c->ev[0], c->stream));
     if (kop == OP_INTERSECT) {
         launch_tiles<OP_INTERSECT, MODE_STAGE>(c, b->d_descs, b->d_tpair, b->d_ta0,
                                                T, b->d_stage, stride, b->d_tcnt,
@@ -2328,7 +2329,7 @@ static int batch_tiles_seq(ua_ctx *c, ua_batch *b, int kop, u64 stride) {
         launch_tiles<OP_UNION, MODE_COUNT>(c, b->d_descs, b->d_tpair, b->d_ta0, T,
                                            nullptr, 0, b->d_tcnt, nullptr, nullptr);
     }
-    HIP_TRY(hipEventRecord(c->ev[1], c->stream));
+    if (record_events) HIP_TRY(hipEventRecord(c->ev[1], c->stream));
     return UA_OK;
 }
 
@@ -2336,18 +2337,18 @@ static int batch_tiles_seq(ua_ctx *c, ua_batch *b, int kop, u64 stride) {
  * into a hipGraph for intersect/diff; union's WRITE pass is a second tile
  * kernel that needs live events, so union stays eager throughout) */
 static int batch_tail_seq(ua_ctx *c, ua_batch *b, int kop, u64 stride,
-                          u64 *host_pout) {
+                          u64 *host_pout, bool record_events) {
     u64 T = b->total_tiles;
     hipLaunchKernelGGL(k_scan1, dim3((u32)b->nchunks), dim3(UA_BLOCK), 0, c->stream,
                        b->d_tcnt, T + 1, b->d_toff, b->d_part);
     hipLaunchKernelGGL(k_scan2, dim3(1), dim3(UA_BLOCK), 0, c->stream, b->d_part,
                        b->nchunks);
     if (kop == OP_UNION) {
-        HIP_TRY(hipEventRecord(c->ev[2], c->stream));
+        if (record_events) HIP_TRY(hipEventRecord(c->ev[2], c->stream));
         launch_tiles<OP_UNION, MODE_WRITE>(c, b->d_descs, b->d_tpair, b->d_ta0, T,
                                            nullptr, 0, b->d_tcnt, b->d_toff,
                                            b->d_part);
-        HIP_TRY(hipEventRecord(c->ev[3], c->stream));
+        if (record_events) HIP_TRY(hipEventRecord(c->ev[3], c->stream));
     } else {
         hipLaunchKernelGGL(k_compact, dim3((u32)((T + 15) / 16)), dim3(UA_BLOCK), 0,
                            c->stream, b->d_descs, b->d_tpair, b->d_tcnt, b->d_toff,
@@ -2361,10 +2362,78 @@ static int batch_tail_seq(ua_ctx *c, ua_batch *b, int kop, u64 stride,
     return UA_OK;
 }
 
-extern "C" int ua_batch_run(ua_ctx *c, ua_batch *b, int op, uint64_t *out_lens) {
-    std::lock_guard<std::recursive_mutex> g(c->mu);
+/* one tail dispatch: graph replay when available, else eager enqueue.
+ * Returns UA_OK with the tail enqueued (or already executed, for the one
+ * capture-validation run). */
+static int batch_tail_dispatch(ua_ctx *c, ua_batch *b, int kop, u64 stride,
+                               bool record_events) {
+#if UA_GRAPH
+    if (kop != OP_UNION) { /* union's tail holds an event-timed kernel */
+        if (!b->gexec[kop] && !b->no_graph) {
+            /* capture the tail once and VALIDATE with a launch+sync; any
+             * failure makes this batch permanently eager — self-healing,
+             * never fatal.  The one validation launch IS this run's tail
+             * (capture itself executes nothing). */
+            hipError_t ce = hipStreamBeginCapture(c->stream,
+                                                  hipStreamCaptureModeThreadLocal);
+            if (ce == hipSuccess) {
+                int rc = batch_tail_seq(c, b, kop, stride, b->h_pout, false);
+                hipGraph_t gr = nullptr;
+                hipError_t ee = hipStreamEndCapture(c->stream, &gr);
+                if (rc == UA_OK && ee == hipSuccess && gr) {
+                    hipError_t ie = hipGraphInstantiate(&b->gexec[kop], gr, nullptr,
+                                                        nullptr, 0);
+                    if (ie == hipSuccess) {
+                        hipError_t le = hipGraphLaunch(b->gexec[kop], c->stream);
+                        if (le == hipSuccess) le = hipStreamSynchronize(c->stream);
+                        if (le == hipSuccess) {
+                            (void)hipGraphDestroy(gr);
+                            return UA_OK; /* tail executed */
+                        }
+                        if (getenv("UA_DEBUG"))
+                            fprintf(stderr, "[ua] graph launch failed: %s\n",
+                                    hipGetErrorName(le));
+                        (void)hipGraphExecDestroy(b->gexec[kop]);
+                        b->gexec[kop] = nullptr;
+                    } else if (getenv("UA_DEBUG")) {
+                        fprintf(stderr, "[ua] graph instantiate failed: %s\n",
+                                hipGetErrorName(ie));
+                        b->gexec[kop] = nullptr;
+                    }
+                    (void)hipGraphDestroy(gr);
+                } else {
+                    if (getenv("UA_DEBUG"))
+                        fprintf(stderr, "[ua] capture failed (rc=%d, end=%s)\n", rc,
+                                hipGetErrorName(ee));
+                    if (gr) (void)hipGraphDestroy(gr);
+                }
+                b->no_graph = true;
+                (void)hipGetLastError();
+            } else {
+                b->no_graph = true;
+                (void)hipGetLastError();
+            }
+        } else if (b->gexec[kop]) {
+            hipError_t le = hipGraphLaunch(b->gexec[kop], c->stream);
+            if (le == hipSuccess) return UA_OK;
+            (void)hipGraphExecDestroy(b->gexec[kop]);
+            b->gexec[kop] = nullptr;
+            b->no_graph = true;
+            (void)hipGetLastError();
+        }
+    }
+#endif
+    return batch_tail_seq(c, b, kop, stride, b->h_pout, record_events);
+}
+
+/* n_runs passes of one op over the prepared batch, enqueued back-to-back
+ * with ONE sync at the end (the repeated-query serving shape: no host
+ * round-trip between runs).  Kernel events are sampled on the LAST pass
+ * (stats count it as one sampled launch). */
+static int batch_run_locked(ua_ctx *c, ua_batch *b, int op, int n_runs,
+                            uint64_t *out_lens) {
     HIP_TRY(hipSetDevice(c->device));
-    if (b->n_pairs == 0) return UA_OK;
+    if (b->n_pairs == 0 || n_runs <= 0) return UA_OK;
     int kop = (op == UA_OP_INTERSECT) ? OP_INTERSECT
               : (op == UA_OP_MERGE) ? OP_UNION
               : (op == UA_OP_DIFFERENCE) ? OP_DIFF
@@ -2377,28 +2446,32 @@ extern "C" int ua_batch_run(ua_ctx *c, ua_batch *b, int op, uint64_t *out_lens) 
     }
     const bool lb = (kop == OP_UNION) ? (bool)UA_LOOKBACK_UNION : (bool)UA_LOOKBACK;
     bool two_kernels = false;
+    int rc;
     if (lb) {
-        if (b->lb_gen >= UA_LB_GEN_MAX) {
-            HIP_TRY(hipMemsetAsync(b->d_lbf, 0, (T + 1) * sizeof(u64), c->stream));
-            b->lb_gen = 0;
-        }
-        b->lb_gen += 1;
-        u64 gen = b->lb_gen;
-        HIP_TRY(hipEventRecord(c->ev[0], c->stream));
-        if (kop == OP_INTERSECT) {
-            launch_tiles<OP_INTERSECT, MODE_LOOKBACK>(c, b->d_descs, b->d_tpair,
+        for (int it = 0; it < n_runs; it++) {
+            bool rec = (it == n_runs - 1);
+            if (b->lb_gen >= UA_LB_GEN_MAX) {
+                HIP_TRY(hipMemsetAsync(b->d_lbf, 0, (T + 1) * sizeof(u64), c->stream));
+                b->lb_gen = 0;
+            }
+            b->lb_gen += 1;
+            u64 gen = b->lb_gen;
+            if (rec) HIP_TRY(hipEventRecord(c->ev[0], c->stream));
+            if (kop == OP_INTERSECT) {
+                launch_tiles<OP_INTERSECT, MODE_LOOKBACK>(c, b->d_descs, b->d_tpair,
+                                                          b->d_ta0, T, b->d_lbf, gen,
+                                                          nullptr, b->d_pout, nullptr);
+            } else if (kop == OP_DIFF) {
+                launch_tiles<OP_DIFF, MODE_LOOKBACK>(c, b->d_descs, b->d_tpair,
+                                                     b->d_ta0, T, b->d_lbf, gen,
+                                                     nullptr, b->d_pout, nullptr);
+            } else {
+                launch_tiles<OP_UNION, MODE_LOOKBACK>(c, b->d_descs, b->d_tpair,
                                                       b->d_ta0, T, b->d_lbf, gen,
                                                       nullptr, b->d_pout, nullptr);
-        } else if (kop == OP_DIFF) {
-            launch_tiles<OP_DIFF, MODE_LOOKBACK>(c, b->d_descs, b->d_tpair, b->d_ta0,
-                                                 T, b->d_lbf, gen, nullptr, b->d_pout,
-                                                 nullptr);
-        } else {
-            launch_tiles<OP_UNION, MODE_LOOKBACK>(c, b->d_descs, b->d_tpair, b->d_ta0,
-                                                  T, b->d_lbf, gen, nullptr, b->d_pout,
-                                                  nullptr);
+            }
+            if (rec) HIP_TRY(hipEventRecord(c->ev[1], c->stream));
         }
-        HIP_TRY(hipEventRecord(c->ev[1], c->stream));
         HIP_TRY(hipMemcpyAsync(out_lens, b->d_pout, (size_t)b->n_pairs * sizeof(u64),
                                hipMemcpyDeviceToHost, c->stream));
         HIP_TRY(hipStreamSynchronize(c->stream));
@@ -2423,75 +2496,10 @@ extern "C" int ua_batch_run(ua_ctx *c, ua_batch *b, int op, uint64_t *out_lens) 
                 return UA_ERR_NOMEM;
             }
         }
-        int rc;
-        if ((rc = batch_tiles_seq(c, b, kop, stride))) return rc;
-        bool ran_tail = false;
-#if UA_GRAPH
-        if (kop != OP_UNION) { /* union's tail holds an event-timed kernel */
-            if (!b->gexec[kop] && !b->no_graph) {
-                /* capture the tail once and VALIDATE with a launch+sync; any
-                 * failure makes this batch permanently eager — self-healing,
-                 * never fatal.  The one validation launch IS this run's tail
-                 * (capture itself executes nothing). */
-                hipError_t ce = hipStreamBeginCapture(
-                    c->stream, hipStreamCaptureModeThreadLocal);
-                if (ce == hipSuccess) {
-                    rc = batch_tail_seq(c, b, kop, stride, b->h_pout);
-                    hipGraph_t gr = nullptr;
-                    hipError_t ee = hipStreamEndCapture(c->stream, &gr);
-                    if (rc == UA_OK && ee == hipSuccess && gr) {
-                        hipError_t ie = hipGraphInstantiate(&b->gexec[kop], gr,
-                                                            nullptr, nullptr, 0);
-                        if (ie == hipSuccess) {
-                            hipError_t le = hipGraphLaunch(b->gexec[kop], c->stream);
-                            if (le == hipSuccess)
-                                le = hipStreamSynchronize(c->stream);
-                            if (le == hipSuccess) {
-                                ran_tail = true;
-                            } else {
-                                if (getenv("UA_DEBUG"))
-                                    fprintf(stderr, "[ua] graph launch failed: %s\n",
-                                            hipGetErrorName(le));
-                                (void)hipGraphExecDestroy(b->gexec[kop]);
-                                b->gexec[kop] = nullptr;
-                                b->no_graph = true;
-                                (void)hipGetLastError();
-                            }
-                        } else {
-                            if (getenv("UA_DEBUG"))
-                                fprintf(stderr, "[ua] graph instantiate failed: %s\n",
-                                        hipGetErrorName(ie));
-                            b->gexec[kop] = nullptr;
-                            b->no_graph = true;
-                        }
-                        (void)hipGraphDestroy(gr);
-                    } else {
-                        if (getenv("UA_DEBUG"))
-                            fprintf(stderr, "[ua] capture failed (rc=%d, end=%s)\n",
-                                    rc, hipGetErrorName(ee));
-                        if (gr) (void)hipGraphDestroy(gr);
-                        b->no_graph = true;
-                        (void)hipGetLastError();
-                    }
-                } else {
-                    b->no_graph = true;
-                    (void)hipGetLastError();
-                }
-            } else if (b->gexec[kop]) {
-                hipError_t le = hipGraphLaunch(b->gexec[kop], c->stream);
-                if (le == hipSuccess) {
-                    ran_tail = true;
-                } else {
-                    (void)hipGraphExecDestroy(b->gexec[kop]);
-                    b->gexec[kop] = nullptr;
-                    b->no_graph = true;
-                    (void)hipGetLastError();
-                }
-            }
-        }
-#endif
-        if (!ran_tail) {
-            if ((rc = batch_tail_seq(c, b, kop, stride, b->h_pout))) return rc;
+        for (int it = 0; it < n_runs; it++) {
+            bool rec = (it == n_runs - 1);
+            if ((rc = batch_tiles_seq(c, b, kop, stride, rec))) return rc;
+            if ((rc = batch_tail_dispatch(c, b, kop, stride, rec))) return rc;
         }
         HIP_TRY(hipStreamSynchronize(c->stream));
         HIP_TRY(hipGetLastError());
@@ -2512,8 +2520,20 @@ extern "C" int ua_batch_run(ua_ctx *c, ua_batch *b, int op, uint64_t *out_lens) 
     }
     u64 out_elems = 0;
     for (int p = 0; p < b->n_pairs; p++) out_elems += out_lens[p];
-    c->bytes_algo += b->in_bytes + 8 * out_elems;
+    c->bytes_algo += ((u64)n_runs) * b->in_bytes + ((u64)n_runs) * 8 * out_elems;
     return UA_OK;
+}
+
+extern "C" int ua_batch_run(ua_ctx *c, ua_batch *b, int op, uint64_t *out_lens) {
+    std::lock_guard<std::recursive_mutex> g(c->mu);
+    return batch_run_locked(c, b, op, 1, out_lens);
+}
+
+/* pipelined repeated runs: one sync for n_runs passes */
+extern "C" int ua_batch_run_n(ua_ctx *c, ua_batch *b, int op, int n_runs,
+                              uint64_t *out_lens) {
+    std::lock_guard<std::recursive_mutex> g(c->mu);
+    return batch_run_locked(c, b, op, n_runs, out_lens);
 }
 
 extern "C" int ua_merge_batch_dev(ua_ctx *c, const ua_dpair *pairs, int n_pairs,

@@ -148,6 +148,11 @@ typedef struct ua_batch ua_batch;
 enum { UA_OP_INTERSECT = 0, UA_OP_MERGE = 1, UA_OP_DIFFERENCE = 2 };
 int ua_batch_create(ua_ctx *, const ua_dpair *pairs, int n_pairs, ua_batch **out);
 int ua_batch_run(ua_ctx *, ua_batch *, int op, uint64_t *out_lens);
+/* n_runs passes enqueued back-to-back with ONE sync (the repeated-query
+ * serving shape); out_lens are the final pass's lengths (all passes over an
+ * immutable batch produce identical results). */
+int ua_batch_run_n(ua_ctx *, ua_batch *, int op, int n_runs,
+                   uint64_t *out_lens);
 void ua_batch_destroy(ua_ctx *, ua_batch *);
 /* batched pairwise algo.MergeSorted semantics (dedup union, uidlist.go:448) */
 int ua_merge_batch_dev(ua_ctx *, const ua_dpair *pairs, int n_pairs,

@@ -652,3 +652,27 @@ def test_zipf_extreme_pair_shapes(eng):
             assert lens[i] == want[i].size, (op_name, i)
             got = to_np(outs[i][:lens[i]])
             assert np.array_equal(got, want[i]), (op_name, i)
+
+
+def test_batch_run_n_matches_run(eng):
+    """ua_batch_run_n (pipelined passes, one sync) returns the same lens and
+    outputs as per-call runs, for all three ops."""
+    rng = np.random.default_rng(SEED + 77)
+    us_np = [np.sort(rng.choice(1_000_000, size=n, replace=False)).astype(np.uint64)
+             for n in [5000, 70_000, 1]]
+    vs_np = [np.sort(rng.choice(1_000_000, size=n, replace=False)).astype(np.uint64)
+             for n in [60_000, 3000, 400_000]]
+    us = [to_dev(u) for u in us_np]
+    vs = [to_dev(v) for v in vs_np]
+    outs = [torch.empty(u.numel() + v.numel(), dtype=torch.int64, device="cuda:0")
+            for u, v in zip(us, vs)]
+    b = eng.make_batch(us, vs, outs)
+    from dgraph_amd.algo import OP_INTERSECT, OP_MERGE, OP_DIFFERENCE
+    for op in [OP_INTERSECT, OP_MERGE, OP_DIFFERENCE]:
+        lens1 = b.run(op)
+        snap1 = [to_np(outs[i][:lens1[i]]).copy() for i in range(3)]
+        lens5 = b.run_n(op, 5)
+        assert lens5 == lens1
+        for i in range(3):
+            assert np.array_equal(to_np(outs[i][:lens5[i]]), snap1[i])
+    b.close()
