@@ -33,6 +33,7 @@
 #include <cstdlib>
 #include <cstring>
 #include <mutex>
+#include <unordered_map>
 #include <vector>
 
 #include "../../include/uidalgo.h"
@@ -416,12 +417,12 @@ __device__ __forceinline__ int tile_search(const u64 *As, int alen, const u64 *B
  * the loop body.  As and Bs must live in the SAME LDS array (smem) so the
  * refill address can select between them.  aoff/boff are u64-element
  * offsets of As/Bs within smem. */
-template <int OP>
+template <int OP, int W>
 __device__ __forceinline__ int tile_walk2(const u64 *smembase, int aoff, int alen,
                                           int boff, int blen, u64 a_before, bool has_ab,
                                           u64 b_before, bool has_bb,
                                           bool has_bn, int s0, int s1, int i0,
-                                          u64 (&em)[UA_WPT], u32 &flags) {
+                                          u64 (&em)[W], u32 &flags) {
     const u64 *As = smembase + aoff;
     const u64 *Bs = smembase + boff;
     int i = i0, j = s0 - i0;
@@ -443,7 +444,7 @@ __device__ __forceinline__ int tile_walk2(const u64 *smembase, int aoff, int ale
         prev_out = d_prev_stream(As, i, a_before, has_ab, Bs, j, b_before, has_bb,
                                  has_prev);
 #pragma unroll
-    for (int s = 0; s < UA_WPT; s++) {
+    for (int s = 0; s < W; s++) {
         if (s >= steps) break;
         bool inA = i < alen, inB = j < blen;
         if (!inA && !inB) break;
@@ -1165,6 +1166,310 @@ __global__ __launch_bounds__(UA_BLOCK) void k_tiles_p2(
         __builtin_amdgcn_s_barrier();
         m = mn;
         f = fn;
+    }
+}
+
+/* ---- register-staged single-buffer pipeline (UA_RPIPE, runtime env) ----
+ *
+ * The two round-1 pipeline rejects both HALVED residency (double LDS buffer
+ * -> 4 WGs/CU); this variant keeps ONE tile-sized LDS buffer and stages the
+ * NEXT tile in registers instead: 512-thread persistent workgroups
+ * grid-stride over tiles; each iteration ds_writes the staged registers
+ * into LDS, immediately issues the 16-B global loads for tile t+G, and only
+ * then walks — so the next tile's HBM latency hides under this tile's
+ * search+walk, with LDS residency unchanged.  All barriers are RAW
+ * (s_waitcnt lgkmcnt only): __syncthreads fences global too and would drain
+ * vmcnt, killing the in-flight prefetch (the presumed reason UA_PIPE saw no
+ * overlap).  STAGE/COUNT/WRITE/DIRECT only — LOOKBACK spins on global
+ * atomics whose vmcnt waits would serialize the prefetch, and its
+ * inter-block dependence needs grid <= residency; it keeps k_tiles. */
+
+#define RP_BLOCK 512
+#define RP_WPT (UA_TILE / RP_BLOCK)
+
+/* global-address-space loads: the staged pointers come out of UaDesc (a
+ * value loaded from memory), so hipcc emits FLAT loads for them — and on
+ * gfx9-family flat ops count on lgkmcnt too, which would make every raw
+ * lgkmcnt barrier wait for the in-flight prefetch.  The addrspacecast pins
+ * them to global_load_* (vmcnt only). */
+#define UA_AS_GLOBAL __attribute__((address_space(1)))
+typedef u64 u64x2 __attribute__((ext_vector_type(2)));
+__device__ __forceinline__ u64x2 d_gload16(const u64 *p) {
+    return *(const UA_AS_GLOBAL u64x2 *)(uintptr_t)p;
+}
+__device__ __forceinline__ u64 d_gload8(const u64 *p) {
+    return *(const UA_AS_GLOBAL u64 *)(uintptr_t)p;
+}
+
+/* Issue the global loads for tile m into registers (two 16-B body units +
+ * one lane-designated single: 0=a_before 1=b_before 2=b_next 3=a_head
+ * 4=b_head 5=a_tail 6=b_tail).  Bodies are the 16-B aligned middles (same
+ * parity layout as the glds fill: As = smem+ashift, Bs = smem+boff so LDS
+ * commits are 16-B aligned too).  Bare locals, not a struct: hipcc spills a
+ * loop-carried struct of vectors to scratch.  The single is ONE
+ * address-selected load, not a switch of loads: exec-masked case loads
+ * into one register WAW-serialize on vmcnt(0) drains. */
+__device__ __forceinline__ void d_rp_load(const TileMeta &m, int tid, u64x2 &st0,
+                                          u64x2 &st1, u64 &sng, const u64 *safe) {
+    if (!m.valid) return;
+    int ash = (int)((((uintptr_t)(m.d.u + m.a0)) >> 3) & 1);
+    int bsh = (int)((((uintptr_t)(m.d.v + m.b0)) >> 3) & 1);
+    int ah = ash < m.alen ? ash : m.alen;
+    int bh = bsh < m.blen ? bsh : m.blen;
+    int na2 = (m.alen - ah) >> 1, atail = (m.alen - ah) & 1;
+    int nb2 = (m.blen - bh) >> 1, btail = (m.blen - bh) & 1;
+    /* branchless single load per unit: exec-masked per-branch loads into one
+     * register WAW-serialize on vmcnt drains; out-of-range lanes read `safe`
+     * (a broadcast dummy) instead */
+    {
+        int u = tid;
+        const u64 *p = (u < na2)         ? m.d.u + m.a0 + ah + 2 * u
+                       : (u - na2 < nb2) ? m.d.v + m.b0 + bh + 2 * (u - na2)
+                                         : safe;
+        st0 = d_gload16(p);
+    }
+    {
+        int u = tid + RP_BLOCK;
+        const u64 *p = (u < na2)         ? m.d.u + m.a0 + ah + 2 * u
+                       : (u - na2 < nb2) ? m.d.v + m.b0 + bh + 2 * (u - na2)
+                                         : safe;
+        st1 = d_gload16(p);
+    }
+    const u64 *sp = safe;
+    switch (tid) {
+    case 0: if (m.has_ab) sp = m.d.u + m.a0 - 1; break;
+    case 1: if (m.b0 > 0) sp = m.d.v + m.b0 - 1; break;
+    case 2: if (m.has_bn) sp = m.d.v + m.b0 + m.blen; break;
+    case 3: if (ah) sp = m.d.u + m.a0; break;
+    case 4: if (bh) sp = m.d.v + m.b0; break;
+    case 5: if (atail) sp = m.d.u + m.a0 + m.alen - 1; break;
+    case 6: if (btail) sp = m.d.v + m.b0 + m.blen - 1; break;
+    default: break;
+    }
+    /* opaque barrier on sp: keep hipcc from sinking one load per case */
+    asm volatile("" : "+v"(sp));
+    /* raw value; the not-applicable zeroing happens at COMMIT time so no
+     * instruction here consumes the load (a use would emit a vmcnt drain
+     * in front of the walk and kill the whole prefetch) */
+    sng = d_gload8(sp);
+}
+
+/* ds_write the staged tile into LDS (consumes the staged regs: the
+ * compiler's counted vmcnt waits land here, not at a barrier). */
+__device__ __forceinline__ void d_rp_commit(const TileMeta &m, int tid, u64 *smem,
+                                            u64x2 st0, u64x2 st1, u64 sng,
+                                            int &aoff, int &boff, u64 *s_ab,
+                                            u64 *s_bb) {
+    int ash = (int)((((uintptr_t)(m.d.u + m.a0)) >> 3) & 1);
+    int bsh = (int)((((uintptr_t)(m.d.v + m.b0)) >> 3) & 1);
+    aoff = ash;
+    boff = ((ash + m.alen + 1) & ~1) + bsh;
+    if (!m.valid) return;
+    int ah = ash < m.alen ? ash : m.alen;
+    int bh = bsh < m.blen ? bsh : m.blen;
+    int na2 = (m.alen - ah) >> 1, atail = (m.alen - ah) & 1;
+    int nb2 = (m.blen - bh) >> 1, btail = (m.blen - bh) & 1;
+    {
+        int u = tid;
+        if (u < na2)
+            *(u64x2 *)&smem[aoff + ah + 2 * u] = st0;
+        else if (u - na2 < nb2)
+            *(u64x2 *)&smem[boff + bh + 2 * (u - na2)] = st0;
+    }
+    {
+        int u = tid + RP_BLOCK;
+        if (u < na2)
+            *(u64x2 *)&smem[aoff + ah + 2 * u] = st1;
+        else if (u - na2 < nb2)
+            *(u64x2 *)&smem[boff + bh + 2 * (u - na2)] = st1;
+    }
+    /* singles: zero the not-applicable cases here (deferred from load) */
+    switch (tid) {
+    case 0: *s_ab = m.has_ab ? sng : 0; break;
+    case 1: *s_bb = (m.b0 > 0) ? sng : 0; break;
+    case 2: smem[boff + m.blen] = m.has_bn ? sng : 0; break; /* lookahead */
+    case 3: if (ah) smem[aoff] = sng; break;
+    case 4: if (bh) smem[boff] = sng; break;
+    case 5: if (atail) smem[aoff + m.alen - 1] = sng; break;
+    case 6: if (btail) smem[boff + m.blen - 1] = sng; break;
+    default: break;
+    }
+}
+
+/* block scan for NB threads with a RAW barrier (no vmcnt drain) */
+template <int NB>
+__device__ __forceinline__ void d_block_scan_rawN(int tid, u32 cnt, u32 *wsum,
+                                                  u32 &excl, u32 &total) {
+    int lane = tid & 63, wv = tid >> 6;
+    u32 incl = cnt;
+#pragma unroll
+    for (int o = 1; o < 64; o <<= 1) {
+        u32 x = __shfl_up(incl, o);
+        if (lane >= o) incl += x;
+    }
+    if (lane == 63) wsum[wv] = incl;
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+    u32 wbase = 0, tot = 0;
+#pragma unroll
+    for (int w = 0; w < NB / 64; w++) {
+        u32 sv = wsum[w];
+        if (w < wv) wbase += sv;
+        tot += sv;
+    }
+    total = tot;
+    excl = wbase + incl - cnt;
+}
+
+/* d_tile_meta with the pair descriptor cached from the previous
+ * (consecutive) tile: descs[p] is a 3-line struct load that would otherwise
+ * sit as a serial L2 round trip at every loop head. */
+__device__ __forceinline__ TileMeta d_tile_meta_seq(const UaDesc *__restrict__ descs,
+                                                    const u32 *__restrict__ tile_pair,
+                                                    const u32 *__restrict__ tile_a0,
+                                                    u64 total_tiles, u64 t, u64 tend,
+                                                    const TileMeta &prev) {
+    TileMeta m;
+    m.valid = (t < tend);
+    if (!m.valid) {
+        m.alen = m.blen = 0;
+        m.a0 = m.b0 = 0;
+        m.has_ab = m.has_bn = false;
+        return m;
+    }
+    m.p = tile_pair[t];
+    if (prev.valid && prev.p == m.p) m.d = prev.d;
+    else m.d = descs[m.p];
+    u64 lt = t - m.d.tile_base;
+    u64 path = m.d.n + m.d.m;
+    u64 d0 = lt * UA_TILE;
+    if (d0 > path) d0 = path;
+    u64 d1 = d0 + UA_TILE;
+    if (d1 > path) d1 = path;
+    u32 a1 = (t + 1 < total_tiles && tile_pair[t + 1] == m.p) ? tile_a0[t + 1]
+                                                              : (u32)m.d.n;
+    m.a0 = tile_a0[t];
+    m.b0 = (u32)(d0 - m.a0);
+    u32 b1 = (u32)(d1 - a1);
+    m.alen = (int)(a1 - m.a0);
+    m.blen = (int)(b1 - m.b0);
+    m.has_ab = (m.a0 > 0);
+    m.has_bn = ((u64)b1 < m.d.m);
+    return m;
+}
+
+template <int OP, int MODE>
+__global__ __launch_bounds__(RP_BLOCK, 8) void k_tiles_rp(
+    const UaDesc *__restrict__ descs, const u32 *__restrict__ tile_pair,
+    const u32 *__restrict__ tile_a0, u64 total_tiles,
+    u64 *__restrict__ staging, u64 stage_stride, u32 *__restrict__ tile_cnt,
+    const u64 *__restrict__ offs, const u64 *__restrict__ partials) {
+    __shared__ __align__(16) u64 smem[UA_TILE + 4];
+    __shared__ u32 wsum[RP_BLOCK / 64];
+    __shared__ u64 s_ab, s_bb;
+
+    int tid = threadIdx.x;
+    /* contiguous chunk per workgroup: consecutive tiles keep the meta words
+     * (tile_pair/tile_a0, u32) L1-hot and the pair descriptor reg-cached */
+    u64 nb = gridDim.x;
+    u64 len = (total_tiles + nb - 1) / nb;
+    u64 t = (u64)blockIdx.x * len;
+    u64 tend = t + len;
+    if (tend > total_tiles) tend = total_tiles;
+    if (t >= tend) return;
+
+    TileMeta mz;
+    mz.valid = false;
+    TileMeta m = d_tile_meta_seq(descs, tile_pair, tile_a0, total_tiles, t, tend, mz);
+    u64x2 st0 = {}, st1 = {};
+    u64 sng = 0;
+    const u64 *safe = (const u64 *)descs;
+    d_rp_load(m, tid, st0, st1, sng, safe);
+
+    for (; t < tend; t++) {
+        TileMeta mn =
+            d_tile_meta_seq(descs, tile_pair, tile_a0, total_tiles, t + 1, tend, m);
+        int aoff, boff;
+        d_rp_commit(m, tid, smem, st0, st1, sng, aoff, boff, &s_ab, &s_bb);
+        /* pin the commit's stores BEFORE the next tile's load issue: left
+         * free, the scheduler moves the singles ds_writes below the next
+         * sng load sharing their register, and the resulting vmcnt(0)
+         * drains stall wave 0 (and through the barrier, the whole WG) on
+         * the prefetch */
+        __builtin_amdgcn_sched_barrier(0);
+        u64x2 nst0 = {}, nst1 = {};
+        u64 nsng = 0;
+        if (mn.valid) d_rp_load(mn, tid, nst0, nst1, nsng, safe); /* in flight across the walk */
+        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+        __builtin_amdgcn_s_barrier();
+
+        int alen = m.alen, blen = m.blen;
+        int tilelen = alen + blen;
+        int s0 = tid * RP_WPT;
+        int s1 = s0 + RP_WPT;
+        if (s0 > tilelen) s0 = tilelen;
+        if (s1 > tilelen) s1 = tilelen;
+        u64 em[RP_WPT];
+        u32 flags = 0;
+        int cnt = 0;
+        {
+            const u64 *As = smem + aoff, *Bs = smem + boff;
+            int i0 = d_merge_path_lds(As, alen, Bs, blen, s0);
+            cnt = tile_walk2<OP>(smem, aoff, alen, boff, blen, s_ab, m.has_ab, s_bb,
+                                 m.b0 > 0, m.has_bn, s0, s1, i0, em, flags);
+        }
+        if (MODE == MODE_DIRECT) {
+            u64 lt = t - m.d.tile_base;
+            u64 *dst = m.d.out + lt * UA_TILE + (u64)s0;
+            int steps = s1 - s0;
+#pragma unroll
+            for (int q = 0; q < RP_WPT; q++)
+                if (q < steps) dst[q] = em[q];
+        } else {
+            u32 excl, total;
+            d_block_scan_rawN<RP_BLOCK>(tid, (u32)cnt, wsum, excl, total);
+            if (MODE == MODE_COUNT) {
+                if (tid == 0) tile_cnt[t] = total;
+            } else {
+                u64 *dst;
+                u32 lim = (u32)cnt;
+                if (MODE == MODE_STAGE) {
+                    /* stride clamp: invalid (duplicate/unsorted) inputs may
+                     * over-emit; keep writes in bounds */
+                    u32 c0 = excl < (u32)stage_stride ? (u32)stage_stride - excl : 0;
+                    if (lim > c0) lim = c0;
+                    dst = staging + t * stage_stride + excl;
+                    if (tid == 0)
+                        tile_cnt[t] = total < (u32)stage_stride ? total
+                                                                : (u32)stage_stride;
+                } else { /* MODE_WRITE: consumed only after the walk, so the
+                            d_off loads draining the prefetch costs little */
+                    dst = m.d.out +
+                          (d_off(offs, partials, t) - d_off(offs, partials, m.d.tile_base)) +
+                          excl;
+                }
+                if (cnt > 0) {
+                    u32 k = 0;
+#pragma unroll
+                    for (int q = 0; q < RP_WPT; q++) {
+                        if (flags & (1u << q)) {
+                            if (k < lim) dst[k] = em[q];
+                            k++;
+                        }
+                    }
+                }
+            }
+        }
+        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+        __builtin_amdgcn_s_barrier(); /* LDS reads done before the next commit */
+        /* keep the register shift BELOW the walk: its copies consume the
+         * prefetched values, and hoisted above the barrier they would drain
+         * vmcnt before the walk ever starts */
+        __builtin_amdgcn_sched_barrier(0);
+        m = mn;
+        st0 = nst0;
+        st1 = nst1;
+        sng = nsng;
     }
 }
 
@@ -1961,12 +2266,59 @@ extern "C" int ua_stats_get(ua_ctx *c, uint64_t *n_launches, double *kernel_ms,
     return UA_OK;
 }
 
+/* UA_RPIPE=0 disables the register-staged pipelined tile kernel (see
+ * k_tiles_rp); default on for STAGE/COUNT/WRITE/DIRECT. */
+static int rp_enabled() {
+    static int v = -1;
+    if (v < 0) {
+        const char *e = getenv("UA_RPIPE");
+        v = (e && e[0]) ? (e[0] != '0') : 0;
+    }
+    return v;
+}
+
+/* persistent grid for k_tiles_rp: residency blocks (occupancy API x CUs).
+ * Oversizing is harmless here (no inter-block deps in these modes). */
+static u32 rp_grid(const void *kfn, u64 T) {
+    static std::mutex mu;
+    static std::unordered_map<const void *, u32> cache;
+    u32 g;
+    {
+        std::lock_guard<std::mutex> lk(mu);
+        auto it = cache.find(kfn);
+        if (it != cache.end()) {
+            g = it->second;
+        } else {
+            int nb = 0, ndev = 0, cus = 256;
+            hipOccupancyMaxActiveBlocksPerMultiprocessor(&nb, kfn, RP_BLOCK, 0);
+            if (hipGetDevice(&ndev) == hipSuccess) {
+                hipDeviceProp_t p;
+                if (hipGetDeviceProperties(&p, ndev) == hipSuccess)
+                    cus = p.multiProcessorCount;
+            }
+            if (nb < 1) nb = 1;
+            g = (u32)nb * (u32)cus;
+            cache.emplace(kfn, g);
+        }
+    }
+    return (u64)g < T ? g : (u32)T;
+}
+
 /* ---- flat scan helper: cnt u32[n] (+1 zero sentinel at n-1 position
  * provided by caller) -> offs u64[n] exclusive scan ---- */
 template <int OP, int MODE>
 static void launch_tiles(ua_ctx *c, const UaDesc *descs, const u32 *tpair,
                          const u32 *ta0, u64 T, u64 *stage, u64 stride, u32 *tcnt,
                          const u64 *offs, const u64 *part) {
+    if constexpr (MODE != MODE_LOOKBACK) {
+        if (rp_enabled()) {
+            u32 G = rp_grid((const void *)k_tiles_rp<OP, MODE>, T);
+            hipLaunchKernelGGL((k_tiles_rp<OP, MODE>), dim3(G), dim3(RP_BLOCK), 0,
+                               c->stream, descs, tpair, ta0, T, stage, stride, tcnt,
+                               offs, part);
+            return;
+        }
+    }
 #if UA_PIPE
     if constexpr (MODE != MODE_DIRECT && MODE != MODE_LOOKBACK) {
         hipLaunchKernelGGL((k_tiles_pipe<OP, MODE>), dim3((u32)((T + 1) / 2)),
