@@ -51,16 +51,24 @@ typedef uint8_t u8;
 #ifndef UA_TILE
 #define UA_TILE 2048
 #endif      /* merge-path elements per tile */
-#define UA_BLOCK 256      /* threads per workgroup */
-#define UA_WPT (UA_TILE / UA_BLOCK) /* path elements per thread */
+#define UA_BLOCK 256      /* threads per workgroup (aux kernels) */
+#ifndef UA_TBLOCK
+#define UA_TBLOCK 256     /* threads per TILE-kernel workgroup: smaller WGs on
+                           * smaller tiles raise independent tiles per CU (the
+                           * 16-blocks/CU cap) — the chip's own latency hiding,
+                           * which measured better than every software pipeline */
+#endif
+#define UA_WPT (UA_TILE / UA_TBLOCK) /* path elements per thread */
 #define UA_SCAN_CHUNK 2048
 #define UA_PKW 4          /* packed-decode waves (blocks) per workgroup */
 #define UA_MAX_BLOCK_UIDS 256
 #define UA_MAX_DELTAS 1092 /* 64 groups x 17 B, padded */
 
-static_assert(UA_TILE % UA_BLOCK == 0, "tile must divide evenly over threads");
-static_assert(UA_TILE / UA_BLOCK >= 1 && UA_TILE / UA_BLOCK <= 32,
+static_assert(UA_TILE % UA_TBLOCK == 0, "tile must divide evenly over threads");
+static_assert(UA_TILE / UA_TBLOCK >= 1 && UA_TILE / UA_TBLOCK <= 32,
               "emission flags are one u32 bit per path step per thread");
+static_assert(UA_TBLOCK % 64 == 0 && UA_TBLOCK >= 64 && UA_TBLOCK <= 1024,
+              "tile kernel is built from whole wavefronts");
 static_assert(UA_BLOCK == 256, "wave geometry (4 waves/WG) is hard-coded in "
                                "scan/compact/packed kernels");
 
@@ -338,6 +346,7 @@ __device__ __forceinline__ int tile_walk(const u64 *As, int alen, const u64 *Bs,
 
 /* block-wide exclusive scan of per-thread counts: wave __shfl scan + one
  * cross-wave combine — 1 barrier instead of Hillis-Steele's 17 */
+template <int NB>
 __device__ __forceinline__ void d_block_scan(int tid, u32 cnt, u32 *wsum,
                                              u32 &excl, u32 &total) {
     int lane = tid & 63, wv = tid >> 6;
@@ -352,7 +361,7 @@ __device__ __forceinline__ void d_block_scan(int tid, u32 cnt, u32 *wsum,
     u32 wbase = 0;
     u32 tot = 0;
 #pragma unroll
-    for (int w = 0; w < UA_BLOCK / 64; w++) {
+    for (int w = 0; w < NB / 64; w++) {
         u32 s = wsum[w];
         if (w < wv) wbase += s;
         tot += s;
@@ -387,8 +396,8 @@ template <int OP>
 __device__ __forceinline__ int tile_search(const u64 *As, int alen, const u64 *Bs,
                                            int blen, bool has_bn, int tid,
                                            u64 (&em)[UA_WPT], u32 &flags) {
-    int lo = (int)(((long)tid * alen) / UA_BLOCK);
-    int hi = (int)(((long)(tid + 1) * alen) / UA_BLOCK);
+    int lo = (int)(((long)tid * alen) / UA_TBLOCK);
+    int hi = (int)(((long)(tid + 1) * alen) / UA_TBLOCK);
     int bext = blen + (has_bn ? 1 : 0);
     int cnt = 0;
     flags = 0;
@@ -588,25 +597,25 @@ __device__ __forceinline__ void d_fill_lds(u64 *dst, const u64 *__restrict__ src
     const ulonglong2 *vs = (const ulonglong2 *)(src + head);
     if ((((uintptr_t)(dst + head)) & 15) == 0) {
         ulonglong2 *vd = (ulonglong2 *)(dst + head);
-        for (int i = tid; i < nvec; i += UA_BLOCK) vd[i] = vs[i];
+        for (int i = tid; i < nvec; i += UA_TBLOCK) vd[i] = vs[i];
     } else {
-        for (int i = tid; i < nvec; i += UA_BLOCK) {
+        for (int i = tid; i < nvec; i += UA_TBLOCK) {
             ulonglong2 x = vs[i];
             dst[head + 2 * i] = x.x;
             dst[head + 2 * i + 1] = x.y;
         }
     }
-    for (int i = head + 2 * nvec + tid; i < len; i += UA_BLOCK) dst[i] = src[i];
+    for (int i = head + 2 * nvec + tid; i < len; i += UA_TBLOCK) dst[i] = src[i];
 }
 
 template <int OP, int MODE>
-__global__ __launch_bounds__(UA_BLOCK) void k_tiles(
+__global__ __launch_bounds__(UA_TBLOCK) void k_tiles(
     const UaDesc *__restrict__ descs, const u32 *__restrict__ tile_pair,
     const u32 *__restrict__ tile_a0, u64 total_tiles,
     u64 *__restrict__ staging, u64 stage_stride, u32 *__restrict__ tile_cnt,
     const u64 *__restrict__ offs, const u64 *__restrict__ partials /* MODE_WRITE only */) {
     __shared__ __align__(16) u64 smem[UA_TILE + 4];
-    __shared__ u32 scan[UA_BLOCK / 64]; /* per-wave totals for d_block_scan */
+    __shared__ u32 scan[UA_TBLOCK / 64]; /* per-wave totals for d_block_scan */
     __shared__ u64 s_abefore;
     __shared__ u64 s_bbefore;
     __shared__ u64 s_run; /* MODE_LOOKBACK: pair-local exclusive prefix */
@@ -655,24 +664,24 @@ __global__ __launch_bounds__(UA_BLOCK) void k_tiles(
         int ahead = ashift; /* elements before the aligned body (0 or 1) */
         if (ahead > alen) ahead = alen;
         int abody = (alen - ahead) & ~127;
-        for (int e = wv4 * 128; e < abody; e += 4 * 128) {
+        for (int e = wv4 * 128; e < abody; e += (UA_TBLOCK / 64) * 128) {
             const u64 *g = d.u + a0 + ahead + e + lane * 2;
             __builtin_amdgcn_global_load_lds((const u32 *)g,
                                              (u32 *)&smem[ashift + ahead + e], 16, 0, 0);
         }
-        for (int i = ahead + abody + tid; i < alen; i += UA_BLOCK)
+        for (int i = ahead + abody + tid; i < alen; i += UA_TBLOCK)
             As[i] = d.u[a0 + i];
         if (tid < ahead) As[tid] = d.u[a0 + tid];
 
         int bhead = bshift;
         if (bhead > blen) bhead = blen;
         int bbody = (blen - bhead) & ~127;
-        for (int e = wv4 * 128; e < bbody; e += 4 * 128) {
+        for (int e = wv4 * 128; e < bbody; e += (UA_TBLOCK / 64) * 128) {
             const u64 *g = d.v + b0 + bhead + e + lane * 2;
             __builtin_amdgcn_global_load_lds((const u32 *)g,
                                              (u32 *)&smem[boff_base + bhead + e], 16, 0, 0);
         }
-        for (int i = bhead + bbody + tid; i < blen; i += UA_BLOCK)
+        for (int i = bhead + bbody + tid; i < blen; i += UA_TBLOCK)
             Bs[i] = d.v[b0 + i];
         if (tid < bhead) Bs[tid] = d.v[b0 + tid];
         /* drain the LDS-DMA before the barrier: the glds writes count on
@@ -759,7 +768,7 @@ __global__ __launch_bounds__(UA_BLOCK) void k_tiles(
 #endif
 
     u32 excl, total;
-    d_block_scan(tid, (u32)cnt, scan, excl, total);
+    d_block_scan<UA_TBLOCK>(tid, (u32)cnt, scan, excl, total);
 
     if (MODE == MODE_COUNT) {
         if (tid == 0) tile_cnt[t] = total;
@@ -953,7 +962,7 @@ __device__ __forceinline__ void d_tile_body(const TileMeta &m, u64 t, int tid,
                              m.b0 > 0, m.has_bn, s0, s1, i0, em, flags);
     }
     u32 excl, total;
-    d_block_scan(tid, (u32)cnt, scanbuf, excl, total);
+    d_block_scan<UA_BLOCK>(tid, (u32)cnt, scanbuf, excl, total);
     if (!m.valid) return;
     if (MODE == MODE_COUNT) {
         if (tid == 0) tile_cnt[t] = total;
@@ -1695,7 +1704,7 @@ __global__ __launch_bounds__(UA_BLOCK) void k_filter(
      * is published (a successor may overwrite these addresses) */
     asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
     u32 excl, total;
-    d_block_scan(tid, (u32)cnt, scan, excl, total);
+    d_block_scan<UA_BLOCK>(tid, (u32)cnt, scan, excl, total);
     if (tid == 0) {
         u64 st = (lt == 0) ? 2ull : 1ull;
         __hip_atomic_store(&lbf[t], d_lb_word(gen, st, total), __ATOMIC_RELAXED,
@@ -2290,7 +2299,7 @@ static u32 rp_grid(const void *kfn, u64 T) {
             g = it->second;
         } else {
             int nb = 0, ndev = 0, cus = 256;
-            hipOccupancyMaxActiveBlocksPerMultiprocessor(&nb, kfn, RP_BLOCK, 0);
+            (void)hipOccupancyMaxActiveBlocksPerMultiprocessor(&nb, kfn, RP_BLOCK, 0);
             if (hipGetDevice(&ndev) == hipSuccess) {
                 hipDeviceProp_t p;
                 if (hipGetDeviceProperties(&p, ndev) == hipSuccess)
@@ -2336,7 +2345,7 @@ static void launch_tiles(ua_ctx *c, const UaDesc *descs, const u32 *tpair,
         return;
     }
 #endif
-    hipLaunchKernelGGL((k_tiles<OP, MODE>), dim3((u32)T), dim3(UA_BLOCK), 0, c->stream,
+    hipLaunchKernelGGL((k_tiles<OP, MODE>), dim3((u32)T), dim3(UA_TBLOCK), 0, c->stream,
                        descs, tpair, ta0, T, stage, stride, tcnt, offs, part);
 }
 
