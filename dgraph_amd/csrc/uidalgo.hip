@@ -59,6 +59,20 @@ typedef uint8_t u8;
                            * which measured better than every software pipeline */
 #endif
 #define UA_WPT (UA_TILE / UA_TBLOCK) /* path elements per thread */
+#ifndef UA_PAD32
+#define UA_PAD32 0 /* 1 = skewed LDS layout (+1 u64 pad per 32): kills the
+                    * 4-way ds_read_b64 bank conflicts of the walk/search
+                    * gathers at power-of-2 lane strides (SQ: conflicts were
+                    * 51% of LDS cycles); fill switches to 4B-lane LDS-DMA
+                    * on 32-element chunks */
+#endif
+#if UA_PAD32
+#define UA_PX(x) ((x) + ((x) >> 5)) /* flat logical u64 index -> physical */
+#define UA_SMEMN (UA_TILE + UA_TILE / 32 + 8)
+#else
+#define UA_PX(x) (x)
+#define UA_SMEMN (UA_TILE + 4)
+#endif
 #define UA_SCAN_CHUNK 2048
 #define UA_PKW 4          /* packed-decode waves (blocks) per workgroup */
 #define UA_MAX_BLOCK_UIDS 256
@@ -104,6 +118,19 @@ __device__ __forceinline__ int d_merge_path_lds(const u64 *A, int n, const u64 *
     while (lo < hi) {
         int mid = (lo + hi) >> 1;
         if (A[mid] <= B[diag - 1 - mid]) lo = mid + 1;
+        else hi = mid;
+    }
+    return lo;
+}
+
+/* merge-path probe through the (possibly padded) tile layout */
+__device__ __forceinline__ int d_merge_path_px(const u64 *smem, int aoff, int n,
+                                               int boff, int m, int diag) {
+    int lo = diag > m ? diag - m : 0;
+    int hi = diag < n ? diag : n;
+    while (lo < hi) {
+        int mid = (lo + hi) >> 1;
+        if (smem[UA_PX(aoff + mid)] <= smem[UA_PX(boff + diag - 1 - mid)]) lo = mid + 1;
         else hi = mid;
     }
     return lo;
@@ -432,8 +459,6 @@ __device__ __forceinline__ int tile_walk2(const u64 *smembase, int aoff, int ale
                                           u64 b_before, bool has_bb,
                                           bool has_bn, int s0, int s1, int i0,
                                           u64 (&em)[W], u32 &flags) {
-    const u64 *As = smembase + aoff;
-    const u64 *Bs = smembase + boff;
     int i = i0, j = s0 - i0;
     int cnt = 0;
     flags = 0;
@@ -442,16 +467,22 @@ __device__ __forceinline__ int tile_walk2(const u64 *smembase, int aoff, int ale
     int amax = alen > 0 ? alen - 1 : 0;
     int bmax = blen_ext > 0 ? blen_ext - 1 : 0;
     /* frontier: cur + next per side, clamped loads (garbage guarded by
-     * i/j bound checks in the predicates) */
-    u64 a = As[i < alen ? i : amax];
-    u64 an = As[(i + 1) < alen ? (i + 1) : amax];
-    u64 b = Bs[j < blen_ext ? j : bmax];
-    u64 bn = Bs[(j + 1) < blen_ext ? (j + 1) : bmax];
+     * i/j bound checks in the predicates); all LDS gathers go through
+     * UA_PX (identity unless UA_PAD32) */
+    u64 a = smembase[UA_PX(aoff + (i < alen ? i : amax))];
+    u64 an = smembase[UA_PX(aoff + ((i + 1) < alen ? (i + 1) : amax))];
+    u64 b = smembase[UA_PX(boff + (j < blen_ext ? j : bmax))];
+    u64 bn = smembase[UA_PX(boff + ((j + 1) < blen_ext ? (j + 1) : bmax))];
     bool has_prev = false;
     u64 prev_out = 0;
-    if (OP == OP_UNION)
-        prev_out = d_prev_stream(As, i, a_before, has_ab, Bs, j, b_before, has_bb,
-                                 has_prev);
+    if (OP == OP_UNION) {
+        u64 pa = (i > 0) ? smembase[UA_PX(aoff + i - 1)] : a_before;
+        bool hpa = (i > 0) || has_ab;
+        u64 pb = (j > 0) ? smembase[UA_PX(boff + j - 1)] : b_before;
+        bool hpb = (j > 0) || has_bb;
+        has_prev = hpa || hpb;
+        prev_out = !hpa ? pb : (!hpb ? pa : (pa > pb ? pa : pb));
+    }
 #pragma unroll
     for (int s = 0; s < W; s++) {
         if (s >= steps) break;
@@ -484,7 +515,7 @@ __device__ __forceinline__ int tile_walk2(const u64 *smembase, int aoff, int ale
         int ra = (ni + 1) < alen ? (ni + 1) : amax;
         int rb = (nj + 1) < blen_ext ? (nj + 1) : bmax;
         int raddr = takeA ? (aoff + ra) : (boff + rb);
-        u64 r = smembase[raddr];
+        u64 r = smembase[UA_PX(raddr)];
         a = takeA ? an : a;
         b = takeA ? b : bn;
         an = takeA ? r : an;
@@ -614,7 +645,7 @@ __global__ __launch_bounds__(UA_TBLOCK) void k_tiles(
     const u32 *__restrict__ tile_a0, u64 total_tiles,
     u64 *__restrict__ staging, u64 stage_stride, u32 *__restrict__ tile_cnt,
     const u64 *__restrict__ offs, const u64 *__restrict__ partials /* MODE_WRITE only */) {
-    __shared__ __align__(16) u64 smem[UA_TILE + 4];
+    __shared__ __align__(16) u64 smem[UA_SMEMN];
     __shared__ u32 scan[UA_TBLOCK / 64]; /* per-wave totals for d_block_scan */
     __shared__ u64 s_abefore;
     __shared__ u64 s_bbefore;
@@ -638,6 +669,14 @@ __global__ __launch_bounds__(UA_TBLOCK) void k_tiles(
 #ifndef UA_GLDS
 #define UA_GLDS 1 /* global_load_lds (LDS-DMA) fill: +13% vs the 16B reg-staged fill (4.03 vs 3.58 TB/s on cfg2); 0 = reg-staged */
 #endif
+#if UA_PAD32
+#if !UA_GLDS || UA_SEARCH || UA_WALK2X || (defined(UA_WALK2) && UA_WALK2 == 0)
+#error "UA_PAD32 supports only the default glds + walk2 configuration"
+#endif
+    /* padded layout: 4B-lane DMA has no 16B parity constraint */
+    int aoff = 0;
+    int boff = alen;
+#else
 #if UA_GLDS
     /* LDS bases chosen so each side's 16B-aligned glds body lines up with
      * its source POINTER parity (glds writes wave-uniform base + lane*16;
@@ -652,12 +691,45 @@ __global__ __launch_bounds__(UA_TBLOCK) void k_tiles(
     u64 *As = smem;                          /* 16-B aligned */
     u64 *Bs = smem + ((alen + 1) & ~1);      /* rounded up to even: 16-B aligned */
 #endif
+    int aoff = (int)(As - smem);
+    int boff = (int)(Bs - smem);
+#endif
 
     bool has_ab = (a0 > 0);
     bool has_bb = (b0 > 0);
     bool has_bn = ((u64)b1 < d.m);
 #if UA_ABLATE != 2
-#if UA_GLDS
+#if UA_PAD32
+    {
+        /* 4B-lane LDS-DMA on 32-element flat-aligned chunks: each chunk's
+         * physical 256B LDS run is contiguous (pads fall between chunks) */
+        int wv4 = tid >> 6, lane = tid & 63;
+        constexpr int NW = UA_TBLOCK / 64;
+        int c1 = alen & ~31; /* aoff == 0 is chunk-aligned */
+        for (int f = wv4 * 32; f < c1; f += NW * 32) {
+            const u32 *srcp = (const u32 *)(d.u + a0 + f) + lane;
+            __builtin_amdgcn_global_load_lds(srcp, (u32 *)&smem[UA_PX(f)], 4, 0, 0);
+        }
+        for (int i = c1 + tid; i < alen; i += UA_TBLOCK)
+            smem[UA_PX(i)] = d.u[a0 + i];
+
+        int bfirst = ((boff + 31) & ~31);          /* first aligned flat chunk */
+        int b_lead = bfirst - boff;
+        if (b_lead > blen) b_lead = blen;
+        int bend = (boff + blen) & ~31;            /* flat end of aligned body */
+        for (int f = bfirst + wv4 * 32; f < bend; f += NW * 32) {
+            const u32 *srcp = (const u32 *)(d.v + b0 + (f - boff)) + lane;
+            __builtin_amdgcn_global_load_lds(srcp, (u32 *)&smem[UA_PX(f)], 4, 0, 0);
+        }
+        for (int i = tid; i < b_lead; i += UA_TBLOCK)
+            smem[UA_PX(boff + i)] = d.v[b0 + i];
+        int btail = bend - boff;
+        if (btail < b_lead) btail = b_lead;
+        for (int i = btail + tid; i < blen; i += UA_TBLOCK)
+            smem[UA_PX(boff + i)] = d.v[b0 + i];
+        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    }
+#elif UA_GLDS
     {
         /* per-wave LDS-DMA fill: 128 u64 per wave-call (64 lanes x 16 B) */
         int wv4 = tid >> 6, lane = tid & 63;
@@ -695,7 +767,7 @@ __global__ __launch_bounds__(UA_TBLOCK) void k_tiles(
     if (tid == 0) {
         s_abefore = has_ab ? d.u[a0 - 1] : 0;
         if (OP == OP_UNION) s_bbefore = has_bb ? d.v[b0 - 1] : 0;
-        Bs[blen] = has_bn ? d.v[b1] : 0;
+        smem[UA_PX(boff + blen)] = has_bn ? d.v[b1] : 0;
     }
 #endif
     __syncthreads();
@@ -704,7 +776,7 @@ __global__ __launch_bounds__(UA_TBLOCK) void k_tiles(
 #define UA_SEARCH 0 /* 1 = A-indexed gallop search for intersect/diff; measured 3.2 vs 3.45 TB/s for the merge walk on cfg2 - kept for skewed-ratio experiments */
 #endif
 #if UA_ABLATE == 1 /* fill-only: keep the loads live, skip search+walk */
-    u64 ablate_x = As[tid] + Bs[tid & 127];
+    u64 ablate_x = smem[UA_PX(aoff + tid)] + smem[UA_PX(boff + (tid & 127))];
     asm volatile("" ::"v"(ablate_x));
     u64 em[UA_WPT];
     u32 flags = 0;
@@ -717,20 +789,24 @@ __global__ __launch_bounds__(UA_TBLOCK) void k_tiles(
 #ifndef UA_WALK2
 #define UA_WALK2 1 /* 0 = the branchy register-frontier walk */
 #endif
+#if !UA_PAD32
     if (UA_SEARCH && OP != OP_UNION && OP != OP_MERGE_ALL) {
         cnt = tile_search<OP>(As, alen, Bs, blen, has_bn, tid, em, flags);
-    } else {
+    } else
+#endif
+    {
         int tilelen = alen + blen;
         int s0 = tid * UA_WPT;
         int s1 = s0 + UA_WPT;
         if (s0 > tilelen) s0 = tilelen;
         if (s1 > tilelen) s1 = tilelen;
-        int i0 = d_merge_path_lds(As, alen, Bs, blen, s0);
+        int i0 = d_merge_path_px(smem, aoff, alen, boff, blen, s0);
         u64 a_before = s_abefore;
         u64 b_before = (OP == OP_UNION) ? s_bbefore : 0;
 #ifndef UA_WALK2X
 #define UA_WALK2X 0 /* 1 = dual-chain walk (2 independent gather chains/thread) */
 #endif
+#if !UA_PAD32
         if (UA_WALK2X) {
             int smid = s0 + UA_WPT / 2;
             if (smid > tilelen) smid = tilelen;
@@ -740,13 +816,15 @@ __global__ __launch_bounds__(UA_TBLOCK) void k_tiles(
             cnt = tile_walk2x<OP>(smem, (int)(As - smem), alen, (int)(Bs - smem), blen,
                                   a_before, has_ab, b_before, has_bb, has_bn, s0, s1,
                                   i0, i0b, smid, em, flags);
-        } else if (UA_WALK2) {
-            cnt = tile_walk2<OP>(smem, (int)(As - smem), alen, (int)(Bs - smem), blen,
-                                 a_before, has_ab, b_before, has_bb, has_bn, s0, s1,
-                                 i0, em, flags);
-        } else {
+        } else if (!UA_WALK2) {
             cnt = tile_walk<OP>(As, alen, Bs, blen, a_before, has_ab, b_before,
                                 has_bb, has_bn, s0, s1, i0, em, flags);
+        } else
+#endif
+        {
+            cnt = tile_walk2<OP>(smem, aoff, alen, boff, blen,
+                                 a_before, has_ab, b_before, has_bb, has_bn, s0, s1,
+                                 i0, em, flags);
         }
         if (MODE == MODE_DIRECT) {
             /* OP_MERGE_ALL: every path element emits, so the output position
@@ -2278,12 +2356,16 @@ extern "C" int ua_stats_get(ua_ctx *c, uint64_t *n_launches, double *kernel_ms,
 /* UA_RPIPE=0 disables the register-staged pipelined tile kernel (see
  * k_tiles_rp); default on for STAGE/COUNT/WRITE/DIRECT. */
 static int rp_enabled() {
+#if UA_PAD32
+    return 0; /* k_tiles_rp commits an unpadded layout */
+#else
     static int v = -1;
     if (v < 0) {
         const char *e = getenv("UA_RPIPE");
         v = (e && e[0]) ? (e[0] != '0') : 0;
     }
     return v;
+#endif
 }
 
 /* persistent grid for k_tiles_rp: residency blocks (occupancy API x CUs).
