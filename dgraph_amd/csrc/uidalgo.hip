@@ -123,11 +123,31 @@ __device__ __forceinline__ int d_merge_path_lds(const u64 *A, int n, const u64 *
     return lo;
 }
 
+#ifndef UA_NARY
+#define UA_NARY 1 /* 4-ary merge-path probe: 3 independent probe pairs per
+                   * round (one lgkm wait) halve the serial round count of
+                   * the per-thread diagonal search — the search is the
+                   * longest latency chain in the walk phase */
+#endif
+
 /* merge-path probe through the (possibly padded) tile layout */
 __device__ __forceinline__ int d_merge_path_px(const u64 *smem, int aoff, int n,
                                                int boff, int m, int diag) {
     int lo = diag > m ? diag - m : 0;
     int hi = diag < n ? diag : n;
+#if UA_NARY
+    while (hi - lo > 3) {
+        int q = (hi - lo) >> 2;
+        int m1 = lo + q, m2 = lo + 2 * q, m3 = lo + 3 * q;
+        /* all six gathers issue before one wait */
+        u64 a1 = smem[UA_PX(aoff + m1)], b1 = smem[UA_PX(boff + diag - 1 - m1)];
+        u64 a2 = smem[UA_PX(aoff + m2)], b2 = smem[UA_PX(boff + diag - 1 - m2)];
+        u64 a3 = smem[UA_PX(aoff + m3)], b3 = smem[UA_PX(boff + diag - 1 - m3)];
+        bool p1 = a1 <= b1, p2 = a2 <= b2, p3 = a3 <= b3;
+        lo = p3 ? m3 + 1 : (p2 ? m2 + 1 : (p1 ? m1 + 1 : lo));
+        hi = p1 ? (p2 ? (p3 ? hi : m3) : m2) : m1;
+    }
+#endif
     while (lo < hi) {
         int mid = (lo + hi) >> 1;
         if (smem[UA_PX(aoff + mid)] <= smem[UA_PX(boff + diag - 1 - mid)]) lo = mid + 1;
@@ -526,6 +546,57 @@ __device__ __forceinline__ int tile_walk2(const u64 *smembase, int aoff, int ale
     return cnt;
 }
 
+/* INTERSECT/DIFF-only walk with NO value array: emissions are always
+ * A-values, so instead of maintaining em[W] (whose per-step conditional
+ * writes and register shuffling dominate the issue-bound walk loop), track
+ * one takeA bitmask and reconstruct the few matched values from LDS after
+ * the scan: A-index at step s = i0 + popcount(amask below s).  ~0.04
+ * matches/thread on the 1%-overlap headline shape. */
+template <int OP, int W>
+__device__ __forceinline__ int tile_walk3(const u64 *smembase, int aoff, int alen,
+                                          int boff, int blen, bool has_bn,
+                                          int s0, int s1, int i0,
+                                          u32 &flags, u32 &amask) {
+    static_assert(W <= 32, "masks are u32");
+    int i = i0, j = s0 - i0;
+    int cnt = 0;
+    flags = 0;
+    amask = 0;
+    int steps = s1 - s0;
+    int blen_ext = blen + (has_bn ? 1 : 0);
+    int amax = alen > 0 ? alen - 1 : 0;
+    int bmax = blen_ext > 0 ? blen_ext - 1 : 0;
+    u64 a = smembase[UA_PX(aoff + (i < alen ? i : amax))];
+    u64 an = smembase[UA_PX(aoff + ((i + 1) < alen ? (i + 1) : amax))];
+    u64 b = smembase[UA_PX(boff + (j < blen_ext ? j : bmax))];
+    u64 bn = smembase[UA_PX(boff + ((j + 1) < blen_ext ? (j + 1) : bmax))];
+#pragma unroll
+    for (int s = 0; s < W; s++) {
+        if (s >= steps) break;
+        bool inA = i < alen, inB = j < blen;
+        if (!inA && !inB) break;
+        bool takeA = inA && (!inB || a <= b);
+        bool eq = (a == b) && (j < blen_ext);
+        bool emit = (OP == OP_INTERSECT) ? (takeA && eq) : (takeA && !eq);
+        flags |= ((u32)emit) << s;
+        amask |= ((u32)takeA) << s;
+        cnt += emit;
+        int ni = i + (takeA ? 1 : 0);
+        int nj = j + (takeA ? 0 : 1);
+        int ra = (ni + 1) < alen ? (ni + 1) : amax;
+        int rb = (nj + 1) < blen_ext ? (nj + 1) : bmax;
+        int raddr = takeA ? (aoff + ra) : (boff + rb);
+        u64 r = smembase[UA_PX(raddr)];
+        a = takeA ? an : a;
+        b = takeA ? b : bn;
+        an = takeA ? r : an;
+        bn = takeA ? bn : r;
+        i = ni;
+        j = nj;
+    }
+    return cnt;
+}
+
 /* Dual-chain walk (UA_WALK2X): each thread walks TWO independent
  * half-segments (diagonals s0 and s0+W/2) with interleaved steps — two
  * independent LDS-gather dependency chains per thread to close the
@@ -780,14 +851,24 @@ __global__ __launch_bounds__(UA_TBLOCK) void k_tiles(
     asm volatile("" ::"v"(ablate_x));
     u64 em[UA_WPT];
     u32 flags = 0;
+    u32 amask = 0;
+    int w_i0 = 0;
     int cnt = 0;
     (void)s_abefore;
 #else
     u64 em[UA_WPT];
     u32 flags;
+    u32 amask = 0; /* walk3 (intersect/diff): takeA bits for value reconstruction */
+    int w_i0 = 0;
     int cnt;
 #ifndef UA_WALK2
 #define UA_WALK2 1 /* 0 = the branchy register-frontier walk */
+#endif
+#ifndef UA_WALK3
+#define UA_WALK3 1 /* 1 = maskless A-value walk for intersect/diff (no em[] bookkeeping) */
+#endif
+#if UA_WALK3 && UA_SEARCH
+#error "UA_SEARCH fills em[]; build it with -DUA_WALK3=0"
 #endif
 #if !UA_PAD32
     if (UA_SEARCH && OP != OP_UNION && OP != OP_MERGE_ALL) {
@@ -821,7 +902,11 @@ __global__ __launch_bounds__(UA_TBLOCK) void k_tiles(
                                 has_bb, has_bn, s0, s1, i0, em, flags);
         } else
 #endif
-        {
+        if (UA_WALK3 && (OP == OP_INTERSECT || OP == OP_DIFF)) {
+            cnt = tile_walk3<OP, UA_WPT>(smem, aoff, alen, boff, blen, has_bn,
+                                         s0, s1, i0, flags, amask);
+            w_i0 = i0;
+        } else {
             cnt = tile_walk2<OP>(smem, aoff, alen, boff, blen,
                                  a_before, has_ab, b_before, has_bb, has_bn, s0, s1,
                                  i0, em, flags);
@@ -888,12 +973,24 @@ __global__ __launch_bounds__(UA_TBLOCK) void k_tiles(
             u64 *dst = d.out + gbase;
             u64 room = cap - gbase;
             int lim = (int)((room < (u64)cnt) ? room : (u64)cnt);
-            int k = 0;
+            if (UA_WALK3 && (OP == OP_INTERSECT || OP == OP_DIFF)) {
+                /* reconstruct emitted A-values from the takeA mask */
+                u32 f = flags;
+                int k = 0;
+                while (f && k < lim) {
+                    int s = __builtin_ctz(f);
+                    f &= f - 1;
+                    int idx = w_i0 + __popc(amask & ((1u << s) - 1));
+                    dst[k++] = smem[UA_PX(aoff + idx)];
+                }
+            } else {
+                int k = 0;
 #pragma unroll
-            for (int s = 0; s < UA_WPT; s++) {
-                if (flags & (1u << s)) {
-                    if (k < lim) dst[k] = em[s];
-                    k++;
+                for (int s = 0; s < UA_WPT; s++) {
+                    if (flags & (1u << s)) {
+                        if (k < lim) dst[k] = em[s];
+                        k++;
+                    }
                 }
             }
         }
@@ -922,12 +1019,24 @@ __global__ __launch_bounds__(UA_TBLOCK) void k_tiles(
               (d_off(offs, partials, t) - d_off(offs, partials, d.tile_base)) + excl;
     }
     if (cnt > 0) {
-        u32 k = 0;
+        if (UA_WALK3 && (OP == OP_INTERSECT || OP == OP_DIFF)) {
+            /* reconstruct emitted A-values from the takeA mask */
+            u32 f = flags;
+            u32 k = 0;
+            while (f && k < lim) {
+                int s = __builtin_ctz(f);
+                f &= f - 1;
+                int idx = w_i0 + __popc(amask & ((1u << s) - 1));
+                dst[k++] = smem[UA_PX(aoff + idx)];
+            }
+        } else {
+            u32 k = 0;
 #pragma unroll
-        for (int s = 0; s < UA_WPT; s++) {
-            if (flags & (1u << s)) {
-                if (k < lim) dst[k] = em[s];
-                k++;
+            for (int s = 0; s < UA_WPT; s++) {
+                if (flags & (1u << s)) {
+                    if (k < lim) dst[k] = em[s];
+                    k++;
+                }
             }
         }
     }
