@@ -124,10 +124,10 @@ __device__ __forceinline__ int d_merge_path_lds(const u64 *A, int n, const u64 *
 }
 
 #ifndef UA_NARY
-#define UA_NARY 1 /* 4-ary merge-path probe: 3 independent probe pairs per
-                   * round (one lgkm wait) halve the serial round count of
-                   * the per-thread diagonal search — the search is the
-                   * longest latency chain in the walk phase */
+#define UA_NARY 0 /* MEASURED REJECT (0.735 vs 0.710 ms on cfg2): 3 probe
+                   * pairs per round (one lgkm wait) halve the search rounds,
+                   * but the extra probe issue outweighs the latency saved —
+                   * the phase is issue-bound after walk3 */
 #endif
 
 /* merge-path probe through the (possibly padded) tile layout */
