@@ -123,6 +123,10 @@ __device__ __forceinline__ int d_merge_path_lds(const u64 *A, int n, const u64 *
     return lo;
 }
 
+#ifndef UA_WALK3
+#define UA_WALK3 1 /* 1 = maskless A-value walk for intersect/diff (no em[] bookkeeping) */
+#endif
+
 #ifndef UA_NARY
 #define UA_NARY 0 /* MEASURED REJECT (0.735 vs 0.710 ms on cfg2): 3 probe
                    * pairs per round (one lgkm wait) halve the search rounds,
@@ -863,12 +867,6 @@ __global__ __launch_bounds__(UA_TBLOCK) void k_tiles(
     int cnt;
 #ifndef UA_WALK2
 #define UA_WALK2 1 /* 0 = the branchy register-frontier walk */
-#endif
-#ifndef UA_WALK3
-#define UA_WALK3 1 /* 1 = maskless A-value walk for intersect/diff (no em[] bookkeeping) */
-#endif
-#if UA_WALK3 && UA_SEARCH
-#error "UA_SEARCH fills em[]; build it with -DUA_WALK3=0"
 #endif
 #if !UA_PAD32
     if (UA_SEARCH && OP != OP_UNION && OP != OP_MERGE_ALL) {
