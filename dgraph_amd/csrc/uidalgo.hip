@@ -850,6 +850,9 @@ __global__ __launch_bounds__(UA_TBLOCK) void k_tiles(
 #ifndef UA_SEARCH
 #define UA_SEARCH 0 /* 1 = A-indexed gallop search for intersect/diff; measured 3.2 vs 3.45 TB/s for the merge walk on cfg2 - kept for skewed-ratio experiments */
 #endif
+#if UA_SEARCH && UA_WALK3
+#error "UA_SEARCH fills em[]; build it with -DUA_WALK3=0"
+#endif
 #if UA_ABLATE == 1 /* fill-only: keep the loads live, skip search+walk */
     u64 ablate_x = smem[UA_PX(aoff + tid)] + smem[UA_PX(boff + (tid & 127))];
     asm volatile("" ::"v"(ablate_x));
