@@ -1670,6 +1670,135 @@ __global__ __launch_bounds__(RP_BLOCK, 8) void k_tiles_rp(
     }
 }
 
+
+/* ---- double-buffered glds pipeline at FULL occupancy (UA_PP, runtime env;
+ * build with -DUA_TILE=1024 so two buffers fit the round-1 LDS budget) ----
+ *
+ * PIPE2's structure (issue the next tile's LDS-DMA before walking the
+ * current, one vmcnt(0) at the swap) was rejected in round 1 at 4-wave WGs
+ * = 16 waves/CU.  At UA_TILE=1024 both buffers fit in 16.4 KB, so 8 WGs x
+ * 4 waves keep the full 32 waves/CU AND every WG always has a fill in
+ * flight.  This attacks the measured convoy loss: fill-only 0.474 ms,
+ * walk-only 0.476, but fill+walk 0.652 — the equal-share HBM service
+ * synchronizes fill completions across WGs, so whole CUs alternate
+ * all-fill / all-walk phases and HBM idles ~27% of the time.
+ * STAGE/COUNT/WRITE/DIRECT only (LOOKBACK keeps k_tiles: its resolve spins
+ * on global atomics whose waits would drain the in-flight DMA). */
+template <int OP, int MODE>
+__global__ __launch_bounds__(UA_TBLOCK, 8) void k_tiles_pp(
+    const UaDesc *__restrict__ descs, const u32 *__restrict__ tile_pair,
+    const u32 *__restrict__ tile_a0, u64 total_tiles,
+    u64 *__restrict__ staging, u64 stage_stride, u32 *__restrict__ tile_cnt,
+    const u64 *__restrict__ offs, const u64 *__restrict__ partials) {
+    static_assert(UA_TBLOCK == UA_BLOCK, "d_p2_issue strides are 4-wave");
+    __shared__ __align__(16) u64 smem[2][UA_TILE + 4];
+    __shared__ u32 scanb[UA_TBLOCK / 64];
+    __shared__ u64 s_ab[2], s_bb[2];
+
+    int tid = threadIdx.x;
+    u64 nb = gridDim.x;
+    u64 len = (total_tiles + nb - 1) / nb;
+    u64 t = (u64)blockIdx.x * len;
+    u64 tend = t + len;
+    if (tend > total_tiles) tend = total_tiles;
+    if (t >= tend) return;
+
+    TileMeta mz;
+    mz.valid = false;
+    TileMeta m = d_tile_meta_seq(descs, tile_pair, tile_a0, total_tiles, t, tend, mz);
+    P2Fill f = d_p2_issue(m, smem[0], tid, &s_ab[0], &s_bb[0], OP == OP_UNION);
+    int cur = 0;
+    asm volatile("s_waitcnt vmcnt(0) lgkmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+
+    for (; t < tend; t++, cur ^= 1) {
+        TileMeta mn =
+            d_tile_meta_seq(descs, tile_pair, tile_a0, total_tiles, t + 1, tend, m);
+        P2Fill fn{0, 0};
+        if (mn.valid)
+            fn = d_p2_issue(mn, smem[cur ^ 1], tid, &s_ab[cur ^ 1], &s_bb[cur ^ 1],
+                            OP == OP_UNION); /* DMA in flight across the walk */
+
+        const u64 *buf = smem[cur];
+        int alen = m.alen, blen = m.blen;
+        int tilelen = alen + blen;
+        int s0 = tid * UA_WPT;
+        int s1 = s0 + UA_WPT;
+        if (s0 > tilelen) s0 = tilelen;
+        if (s1 > tilelen) s1 = tilelen;
+        u64 em[UA_WPT];
+        u32 flags = 0, amask = 0;
+        int cnt = 0, w_i0 = 0;
+        int i0 = d_merge_path_px(buf, f.aoff, alen, f.boff, blen, s0);
+        if (UA_WALK3 && (OP == OP_INTERSECT || OP == OP_DIFF)) {
+            cnt = tile_walk3<OP, UA_WPT>(buf, f.aoff, alen, f.boff, blen, m.has_bn,
+                                         s0, s1, i0, flags, amask);
+            w_i0 = i0;
+        } else {
+            cnt = tile_walk2<OP>(buf, f.aoff, alen, f.boff, blen, s_ab[cur], m.has_ab,
+                                 s_bb[cur], m.b0 > 0, m.has_bn, s0, s1, i0, em, flags);
+        }
+        if (MODE == MODE_DIRECT) {
+            u64 lt = t - m.d.tile_base;
+            u64 *dst = m.d.out + lt * UA_TILE + (u64)s0;
+            int steps = s1 - s0;
+#pragma unroll
+            for (int q = 0; q < UA_WPT; q++)
+                if (q < steps) dst[q] = em[q];
+        } else {
+            u32 excl, total;
+            d_block_scan_rawN<UA_TBLOCK>(tid, (u32)cnt, scanb, excl, total);
+            if (MODE == MODE_COUNT) {
+                if (tid == 0) tile_cnt[t] = total;
+            } else {
+                u64 *dst;
+                u32 lim = (u32)cnt;
+                if (MODE == MODE_STAGE) {
+                    /* stride clamp as in k_tiles */
+                    u32 c0 = excl < (u32)stage_stride ? (u32)stage_stride - excl : 0;
+                    if (lim > c0) lim = c0;
+                    dst = staging + t * stage_stride + excl;
+                    if (tid == 0)
+                        tile_cnt[t] = total < (u32)stage_stride ? total
+                                                                : (u32)stage_stride;
+                } else { /* MODE_WRITE: post-walk d_off loads may drain the
+                            DMA, but it has already had the walk to land */
+                    dst = m.d.out +
+                          (d_off(offs, partials, t) - d_off(offs, partials, m.d.tile_base)) +
+                          excl;
+                }
+                if (cnt > 0) {
+                    if (UA_WALK3 && (OP == OP_INTERSECT || OP == OP_DIFF)) {
+                        u32 fl = flags;
+                        u32 k = 0;
+                        while (fl && k < lim) {
+                            int s = __builtin_ctz(fl);
+                            fl &= fl - 1;
+                            int idx = w_i0 + __popc(amask & ((1u << s) - 1));
+                            dst[k++] = buf[UA_PX(f.aoff + idx)];
+                        }
+                    } else {
+                        u32 k = 0;
+#pragma unroll
+                        for (int s = 0; s < UA_WPT; s++) {
+                            if (flags & (1u << s)) {
+                                if (k < lim) dst[k] = em[s];
+                                k++;
+                            }
+                        }
+                    }
+                }
+            }
+        }
+        /* drain the next tile's DMA (walk covered its latency) + make sure
+         * every wave is done reading this buffer before it is refilled */
+        asm volatile("s_waitcnt vmcnt(0) lgkmcnt(0)" ::: "memory");
+        __builtin_amdgcn_s_barrier();
+        m = mn;
+        f = fn;
+    }
+}
+
 /* ==================== kernel: bitonic chunk sort (segmented sort stage 1) ====================
  * One workgroup sorts one <=2048-element chunk in LDS (u64 ascending,
  * duplicates kept; padded with UINT64_MAX).  Stage 2 is the merge-path
@@ -2465,6 +2594,15 @@ extern "C" int ua_stats_get(ua_ctx *c, uint64_t *n_launches, double *kernel_ms,
 
 /* UA_RPIPE=0 disables the register-staged pipelined tile kernel (see
  * k_tiles_rp); default on for STAGE/COUNT/WRITE/DIRECT. */
+static int pp_enabled() {
+    static int v = -1;
+    if (v < 0) {
+        const char *e = getenv("UA_PP");
+        v = (e && e[0]) ? (e[0] != '0') : 0;
+    }
+    return v;
+}
+
 static int rp_enabled() {
 #if UA_PAD32
     return 0; /* k_tiles_rp commits an unpadded layout */
@@ -2512,6 +2650,13 @@ static void launch_tiles(ua_ctx *c, const UaDesc *descs, const u32 *tpair,
                          const u32 *ta0, u64 T, u64 *stage, u64 stride, u32 *tcnt,
                          const u64 *offs, const u64 *part) {
     if constexpr (MODE != MODE_LOOKBACK) {
+        if (pp_enabled()) {
+            u32 G = rp_grid((const void *)k_tiles_pp<OP, MODE>, T);
+            hipLaunchKernelGGL((k_tiles_pp<OP, MODE>), dim3(G), dim3(UA_TBLOCK), 0,
+                               c->stream, descs, tpair, ta0, T, stage, stride, tcnt,
+                               offs, part);
+            return;
+        }
         if (rp_enabled()) {
             u32 G = rp_grid((const void *)k_tiles_rp<OP, MODE>, T);
             hipLaunchKernelGGL((k_tiles_rp<OP, MODE>), dim3(G), dim3(RP_BLOCK), 0,
