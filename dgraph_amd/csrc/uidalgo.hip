@@ -550,6 +550,69 @@ __device__ __forceinline__ int tile_walk2(const u64 *smembase, int aoff, int ale
     return cnt;
 }
 
+/* COUNT-only walk: tile_walk2 minus the em[] value writes — MODE_COUNT
+ * needs only the per-thread emission count (union's first pass; walk3
+ * already covers intersect/diff counts).  Dedup state is kept so union
+ * counts match tile_walk2 exactly. */
+template <int OP, int W>
+__device__ __forceinline__ int tile_walk2c(const u64 *smembase, int aoff, int alen,
+                                           int boff, int blen, u64 a_before,
+                                           bool has_ab, u64 b_before, bool has_bb,
+                                           bool has_bn, int s0, int s1, int i0) {
+    int i = i0, j = s0 - i0;
+    int cnt = 0;
+    int steps = s1 - s0;
+    int blen_ext = blen + (has_bn ? 1 : 0);
+    int amax = alen > 0 ? alen - 1 : 0;
+    int bmax = blen_ext > 0 ? blen_ext - 1 : 0;
+    u64 a = smembase[UA_PX(aoff + (i < alen ? i : amax))];
+    u64 an = smembase[UA_PX(aoff + ((i + 1) < alen ? (i + 1) : amax))];
+    u64 b = smembase[UA_PX(boff + (j < blen_ext ? j : bmax))];
+    u64 bn = smembase[UA_PX(boff + ((j + 1) < blen_ext ? (j + 1) : bmax))];
+    bool has_prev = false;
+    u64 prev_out = 0;
+    if (OP == OP_UNION) {
+        u64 pa = (i > 0) ? smembase[UA_PX(aoff + i - 1)] : a_before;
+        bool hpa = (i > 0) || has_ab;
+        u64 pb = (j > 0) ? smembase[UA_PX(boff + j - 1)] : b_before;
+        bool hpb = (j > 0) || has_bb;
+        has_prev = hpa || hpb;
+        prev_out = !hpa ? pb : (!hpb ? pa : (pa > pb ? pa : pb));
+    }
+#pragma unroll
+    for (int s = 0; s < W; s++) {
+        if (s >= steps) break;
+        bool inA = i < alen, inB = j < blen;
+        if (!inA && !inB) break;
+        bool takeA = inA && (!inB || a <= b);
+        bool eq = (a == b) && (j < blen_ext);
+        bool emit;
+        if (OP == OP_INTERSECT) emit = takeA && eq;
+        else if (OP == OP_DIFF) emit = takeA && !eq;
+        else if (OP == OP_MERGE_ALL) emit = true;
+        else {
+            u64 val = takeA ? a : b;
+            emit = !has_prev || val != prev_out;
+            prev_out = val;
+            has_prev = true;
+        }
+        cnt += emit;
+        int ni = i + (takeA ? 1 : 0);
+        int nj = j + (takeA ? 0 : 1);
+        int ra = (ni + 1) < alen ? (ni + 1) : amax;
+        int rb = (nj + 1) < blen_ext ? (nj + 1) : bmax;
+        int raddr = takeA ? (aoff + ra) : (boff + rb);
+        u64 r = smembase[UA_PX(raddr)];
+        a = takeA ? an : a;
+        b = takeA ? b : bn;
+        an = takeA ? r : an;
+        bn = takeA ? bn : r;
+        i = ni;
+        j = nj;
+    }
+    return cnt;
+}
+
 /* INTERSECT/DIFF-only walk with NO value array: emissions are always
  * A-values, so instead of maintaining em[W] (whose per-step conditional
  * writes and register shuffling dominate the issue-bound walk loop), track
@@ -907,6 +970,12 @@ __global__ __launch_bounds__(UA_TBLOCK) void k_tiles(
             cnt = tile_walk3<OP, UA_WPT>(smem, aoff, alen, boff, blen, has_bn,
                                          s0, s1, i0, flags, amask);
             w_i0 = i0;
+        } else if (MODE == MODE_COUNT) {
+            /* count pass needs no values — skip the em[] bookkeeping */
+            cnt = tile_walk2c<OP, UA_WPT>(smem, aoff, alen, boff, blen, a_before,
+                                          has_ab, b_before, has_bb, has_bn, s0, s1,
+                                          i0);
+            flags = 0;
         } else {
             cnt = tile_walk2<OP>(smem, aoff, alen, boff, blen,
                                  a_before, has_ab, b_before, has_bb, has_bn, s0, s1,
