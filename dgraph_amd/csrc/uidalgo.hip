@@ -777,16 +777,20 @@ __device__ __forceinline__ void d_fill_lds(u64 *dst, const u64 *__restrict__ src
     for (int i = head + 2 * nvec + tid; i < len; i += UA_TBLOCK) dst[i] = src[i];
 }
 
-/* one tile's full pipeline (fill -> search/walk -> scan -> emit); factored
- * so k_tiles can amortize per-block fixed costs (preamble, meta chain,
- * dispatch/retire) over UA_TPB consecutive tiles */
 template <int OP, int MODE>
-__device__ __forceinline__ void d_ktile_one(
-    u64 t, int tid, u64 *smem, u32 *scan, u64 &s_abefore, u64 &s_bbefore,
-    u64 &s_run, const UaDesc *__restrict__ descs, const u32 *__restrict__ tile_pair,
-    const u32 *__restrict__ tile_a0, u64 total_tiles, u64 *__restrict__ staging,
-    u64 stage_stride, u32 *__restrict__ tile_cnt, const u64 *__restrict__ offs,
-    const u64 *__restrict__ partials) {
+__global__ __launch_bounds__(UA_TBLOCK) void k_tiles(
+    const UaDesc *__restrict__ descs, const u32 *__restrict__ tile_pair,
+    const u32 *__restrict__ tile_a0, u64 total_tiles,
+    u64 *__restrict__ staging, u64 stage_stride, u32 *__restrict__ tile_cnt,
+    const u64 *__restrict__ offs, const u64 *__restrict__ partials /* MODE_WRITE only */) {
+    __shared__ __align__(16) u64 smem[UA_SMEMN];
+    __shared__ u32 scan[UA_TBLOCK / 64]; /* per-wave totals for d_block_scan */
+    __shared__ u64 s_abefore;
+    __shared__ u64 s_bbefore;
+    __shared__ u64 s_run; /* MODE_LOOKBACK: pair-local exclusive prefix */
+
+    u64 t = blockIdx.x;
+    int tid = threadIdx.x;
     u32 p = tile_pair[t];
     UaDesc d = descs[p];
     u64 lt = t - d.tile_base;
@@ -1105,34 +1109,6 @@ __device__ __forceinline__ void d_ktile_one(
                 }
             }
         }
-    }
-}
-
-#ifndef UA_TPB
-#define UA_TPB 1 /* consecutive tiles per workgroup (amortizes block overhead) */
-#endif
-
-template <int OP, int MODE>
-__global__ __launch_bounds__(UA_TBLOCK) void k_tiles(
-    const UaDesc *__restrict__ descs, const u32 *__restrict__ tile_pair,
-    const u32 *__restrict__ tile_a0, u64 total_tiles,
-    u64 *__restrict__ staging, u64 stage_stride, u32 *__restrict__ tile_cnt,
-    const u64 *__restrict__ offs, const u64 *__restrict__ partials /* MODE_WRITE only */) {
-    __shared__ __align__(16) u64 smem[UA_SMEMN];
-    __shared__ u32 scan[UA_TBLOCK / 64]; /* per-wave totals for d_block_scan */
-    __shared__ u64 s_abefore;
-    __shared__ u64 s_bbefore;
-    __shared__ u64 s_run; /* MODE_LOOKBACK: pair-local exclusive prefix */
-
-    int tid = threadIdx.x;
-    u64 t0 = (u64)blockIdx.x * UA_TPB;
-    u64 t1 = t0 + UA_TPB;
-    if (t1 > total_tiles) t1 = total_tiles;
-    for (u64 t = t0; t < t1; t++) {
-        if (t > t0) __syncthreads(); /* smem reuse: prior tile fully consumed */
-        d_ktile_one<OP, MODE>(t, tid, smem, scan, s_abefore, s_bbefore, s_run,
-                              descs, tile_pair, tile_a0, total_tiles, staging,
-                              stage_stride, tile_cnt, offs, partials);
     }
 }
 
@@ -2775,8 +2751,7 @@ static void launch_tiles(ua_ctx *c, const UaDesc *descs, const u32 *tpair,
         return;
     }
 #endif
-    hipLaunchKernelGGL((k_tiles<OP, MODE>), dim3((u32)((T + UA_TPB - 1) / UA_TPB)),
-                       dim3(UA_TBLOCK), 0, c->stream,
+    hipLaunchKernelGGL((k_tiles<OP, MODE>), dim3((u32)T), dim3(UA_TBLOCK), 0, c->stream,
                        descs, tpair, ta0, T, stage, stride, tcnt, offs, part);
 }
 
