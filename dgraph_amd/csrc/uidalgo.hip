@@ -2506,6 +2506,8 @@ struct ua_ctx {
     void *ws[WS_COUNT] = {};
     size_t ws_cap[WS_COUNT] = {};
     hipEvent_t ev[4] = {};
+    hipStream_t stream_tail = nullptr; /* aux-tail overlap (run_n) */
+    hipEvent_t oev[4] = {}; /* [0..1]=tile_done per set, [2..3]=tail_done */
     /* decoupled-lookback flag array (WS_LBF) bookkeeping */
     uint32_t lbf_gen = 0;
     size_t lbf_cleared = 0; /* bytes of the CURRENT WS_LBF allocation zeroed */
@@ -2545,6 +2547,9 @@ extern "C" int ua_ctx_create(ua_ctx **out, int device) {
          * be captured into a graph on this ROCm (invalid resource handle) */
         HIP_TRY(hipEventRecord(c->ev[i], c->stream));
     }
+    HIP_TRY(hipStreamCreate(&c->stream_tail));
+    for (int i = 0; i < 4; i++)
+        HIP_TRY(hipEventCreateWithFlags(&c->oev[i], hipEventDisableTiming));
     HIP_TRY(hipStreamSynchronize(c->stream));
     *out = c;
     return UA_OK;
@@ -2557,6 +2562,9 @@ extern "C" void ua_ctx_destroy(ua_ctx *c) {
         if (c->ws[i]) (void)hipFree(c->ws[i]);
     for (int i = 0; i < 4; i++)
         if (c->ev[i]) (void)hipEventDestroy(c->ev[i]);
+    for (int i = 0; i < 4; i++)
+        if (c->oev[i]) (void)hipEventDestroy(c->oev[i]);
+    if (c->stream_tail) (void)hipStreamDestroy(c->stream_tail);
     if (c->stream) (void)hipStreamDestroy(c->stream);
     delete c;
 }
@@ -2969,6 +2977,13 @@ struct ua_batch {
     u64 nchunks = 0;
     void *mem = nullptr;     /* one allocation for all metadata arrays */
     u64 *d_stage = nullptr;  /* lazy; stride UA_TILE (fits intersect + diff) */
+    /* set-1 buffers for the overlapped run_n (aux tail of step i on the
+     * tail stream while step i+1's tile kernel fills the other set) */
+    u64 *d_stage1 = nullptr;
+    void *mem1 = nullptr; /* holds d_tcnt1 + d_toff1 + d_part1 */
+    u32 *d_tcnt1 = nullptr;
+    u64 *d_toff1 = nullptr;
+    u64 *d_part1 = nullptr;
     UaDesc *d_descs = nullptr;
     u64 *d_tb = nullptr;
     u32 *d_tpair = nullptr;
@@ -2981,7 +2996,7 @@ struct ua_batch {
     u32 lb_gen = 0;
     /* hipGraph capture of the staged pipeline (per op); out_lens land in the
      * pinned h_pout so the captured D2H copy has a fixed destination */
-    hipGraphExec_t gexec[3] = {};
+    hipGraphExec_t gexec[3][2] = {};
     bool no_graph = false; /* capture/replay failed on this box: stay eager */
     u64 *h_pout = nullptr;
 };
@@ -3070,10 +3085,13 @@ extern "C" void ua_batch_destroy(ua_ctx *c, ua_batch *b) {
     if (!b) return;
     (void)hipSetDevice(c->device);
     for (int i = 0; i < 3; i++)
-        if (b->gexec[i]) (void)hipGraphExecDestroy(b->gexec[i]);
+        for (int s = 0; s < 2; s++)
+            if (b->gexec[i][s]) (void)hipGraphExecDestroy(b->gexec[i][s]);
     if (b->h_pout) (void)hipHostFree(b->h_pout);
     if (b->mem) (void)hipFree(b->mem);
+    if (b->mem1) (void)hipFree(b->mem1);
     if (b->d_stage) (void)hipFree(b->d_stage);
+    if (b->d_stage1) (void)hipFree(b->d_stage1);
     delete b;
 }
 
@@ -3088,16 +3106,18 @@ extern "C" void ua_batch_destroy(ua_ctx *c, ua_batch *b) {
 
 /* the staged TILE kernel (eager, event-timed) */
 static int batch_tiles_seq(ua_ctx *c, ua_batch *b, int kop, u64 stride,
-                           bool record_events) {
+                           bool record_events, int set) {
     u64 T = b->total_tiles;
+    u64 *stage = set ? b->d_stage1 : b->d_stage;
+    u32 *tcnt = set ? b->d_tcnt1 : b->d_tcnt;
     if (record_events) HIP_TRY(hipEventRecord(c->ev[0], c->stream));
     if (kop == OP_INTERSECT) {
         launch_tiles<OP_INTERSECT, MODE_STAGE>(c, b->d_descs, b->d_tpair, b->d_ta0,
-                                               T, b->d_stage, stride, b->d_tcnt,
+                                               T, stage, stride, tcnt,
                                                nullptr, nullptr);
     } else if (kop == OP_DIFF) {
         launch_tiles<OP_DIFF, MODE_STAGE>(c, b->d_descs, b->d_tpair, b->d_ta0, T,
-                                          b->d_stage, stride, b->d_tcnt, nullptr,
+                                          stage, stride, tcnt, nullptr,
                                           nullptr);
     } else {
         launch_tiles<OP_UNION, MODE_COUNT>(c, b->d_descs, b->d_tpair, b->d_ta0, T,
@@ -3111,28 +3131,33 @@ static int batch_tiles_seq(ua_ctx *c, ua_batch *b, int kop, u64 stride,
  * into a hipGraph for intersect/diff; union's WRITE pass is a second tile
  * kernel that needs live events, so union stays eager throughout) */
 static int batch_tail_seq(ua_ctx *c, ua_batch *b, int kop, u64 stride,
-                          u64 *host_pout, bool record_events) {
+                          u64 *host_pout, bool record_events, int set,
+                          hipStream_t st) {
     u64 T = b->total_tiles;
-    hipLaunchKernelGGL(k_scan1, dim3((u32)b->nchunks), dim3(UA_BLOCK), 0, c->stream,
-                       b->d_tcnt, T + 1, b->d_toff, b->d_part);
-    hipLaunchKernelGGL(k_scan2, dim3(1), dim3(UA_BLOCK), 0, c->stream, b->d_part,
+    u32 *tcnt = set ? b->d_tcnt1 : b->d_tcnt;
+    u64 *toff = set ? b->d_toff1 : b->d_toff;
+    u64 *part = set ? b->d_part1 : b->d_part;
+    u64 *stage = set ? b->d_stage1 : b->d_stage;
+    hipLaunchKernelGGL(k_scan1, dim3((u32)b->nchunks), dim3(UA_BLOCK), 0, st,
+                       tcnt, T + 1, toff, part);
+    hipLaunchKernelGGL(k_scan2, dim3(1), dim3(UA_BLOCK), 0, st, part,
                        b->nchunks);
     if (kop == OP_UNION) {
-        if (record_events) HIP_TRY(hipEventRecord(c->ev[2], c->stream));
+        if (record_events) HIP_TRY(hipEventRecord(c->ev[2], st));
         launch_tiles<OP_UNION, MODE_WRITE>(c, b->d_descs, b->d_tpair, b->d_ta0, T,
                                            nullptr, 0, b->d_tcnt, b->d_toff,
                                            b->d_part);
-        if (record_events) HIP_TRY(hipEventRecord(c->ev[3], c->stream));
+        if (record_events) HIP_TRY(hipEventRecord(c->ev[3], st));
     } else {
         hipLaunchKernelGGL(k_compact, dim3((u32)((T + 15) / 16)), dim3(UA_BLOCK), 0,
-                           c->stream, b->d_descs, b->d_tpair, b->d_tcnt, b->d_toff,
-                           b->d_part, b->d_stage, stride, T, kop);
+                           st, b->d_descs, b->d_tpair, tcnt, toff,
+                           part, stage, stride, T, kop);
     }
     u64 poutblk = ((u64)b->n_pairs + UA_BLOCK - 1) / UA_BLOCK;
-    hipLaunchKernelGGL(k_pair_out, dim3((u32)poutblk), dim3(UA_BLOCK), 0, c->stream,
-                       b->d_toff, b->d_part, b->d_tb, b->n_pairs, b->d_pout);
+    hipLaunchKernelGGL(k_pair_out, dim3((u32)poutblk), dim3(UA_BLOCK), 0, st,
+                       toff, part, b->d_tb, b->n_pairs, b->d_pout);
     HIP_TRY(hipMemcpyAsync(host_pout, b->d_pout, (size_t)b->n_pairs * sizeof(u64),
-                           hipMemcpyDeviceToHost, c->stream));
+                           hipMemcpyDeviceToHost, st));
     return UA_OK;
 }
 
@@ -3140,26 +3165,26 @@ static int batch_tail_seq(ua_ctx *c, ua_batch *b, int kop, u64 stride,
  * Returns UA_OK with the tail enqueued (or already executed, for the one
  * capture-validation run). */
 static int batch_tail_dispatch(ua_ctx *c, ua_batch *b, int kop, u64 stride,
-                               bool record_events) {
+                               bool record_events, int set, hipStream_t st) {
 #if UA_GRAPH
     if (kop != OP_UNION) { /* union's tail holds an event-timed kernel */
-        if (!b->gexec[kop] && !b->no_graph) {
+        if (!b->gexec[kop][set] && !b->no_graph) {
             /* capture the tail once and VALIDATE with a launch+sync; any
              * failure makes this batch permanently eager — self-healing,
              * never fatal.  The one validation launch IS this run's tail
              * (capture itself executes nothing). */
-            hipError_t ce = hipStreamBeginCapture(c->stream,
+            hipError_t ce = hipStreamBeginCapture(st,
                                                   hipStreamCaptureModeThreadLocal);
             if (ce == hipSuccess) {
-                int rc = batch_tail_seq(c, b, kop, stride, b->h_pout, false);
+                int rc = batch_tail_seq(c, b, kop, stride, b->h_pout, false, set, st);
                 hipGraph_t gr = nullptr;
-                hipError_t ee = hipStreamEndCapture(c->stream, &gr);
+                hipError_t ee = hipStreamEndCapture(st, &gr);
                 if (rc == UA_OK && ee == hipSuccess && gr) {
-                    hipError_t ie = hipGraphInstantiate(&b->gexec[kop], gr, nullptr,
-                                                        nullptr, 0);
+                    hipError_t ie = hipGraphInstantiate(&b->gexec[kop][set], gr,
+                                                        nullptr, nullptr, 0);
                     if (ie == hipSuccess) {
-                        hipError_t le = hipGraphLaunch(b->gexec[kop], c->stream);
-                        if (le == hipSuccess) le = hipStreamSynchronize(c->stream);
+                        hipError_t le = hipGraphLaunch(b->gexec[kop][set], st);
+                        if (le == hipSuccess) le = hipStreamSynchronize(st);
                         if (le == hipSuccess) {
                             (void)hipGraphDestroy(gr);
                             return UA_OK; /* tail executed */
@@ -3167,12 +3192,12 @@ static int batch_tail_dispatch(ua_ctx *c, ua_batch *b, int kop, u64 stride,
                         if (getenv("UA_DEBUG"))
                             fprintf(stderr, "[ua] graph launch failed: %s\n",
                                     hipGetErrorName(le));
-                        (void)hipGraphExecDestroy(b->gexec[kop]);
-                        b->gexec[kop] = nullptr;
+                        (void)hipGraphExecDestroy(b->gexec[kop][set]);
+                        b->gexec[kop][set] = nullptr;
                     } else if (getenv("UA_DEBUG")) {
                         fprintf(stderr, "[ua] graph instantiate failed: %s\n",
                                 hipGetErrorName(ie));
-                        b->gexec[kop] = nullptr;
+                        b->gexec[kop][set] = nullptr;
                     }
                     (void)hipGraphDestroy(gr);
                 } else {
@@ -3187,17 +3212,17 @@ static int batch_tail_dispatch(ua_ctx *c, ua_batch *b, int kop, u64 stride,
                 b->no_graph = true;
                 (void)hipGetLastError();
             }
-        } else if (b->gexec[kop]) {
-            hipError_t le = hipGraphLaunch(b->gexec[kop], c->stream);
+        } else if (b->gexec[kop][set]) {
+            hipError_t le = hipGraphLaunch(b->gexec[kop][set], st);
             if (le == hipSuccess) return UA_OK;
-            (void)hipGraphExecDestroy(b->gexec[kop]);
-            b->gexec[kop] = nullptr;
+            (void)hipGraphExecDestroy(b->gexec[kop][set]);
+            b->gexec[kop][set] = nullptr;
             b->no_graph = true;
             (void)hipGetLastError();
         }
     }
 #endif
-    return batch_tail_seq(c, b, kop, stride, b->h_pout, record_events);
+    return batch_tail_seq(c, b, kop, stride, b->h_pout, record_events, set, st);
 }
 
 /* n_runs passes of one op over the prepared batch, enqueued back-to-back
@@ -3270,12 +3295,60 @@ static int batch_run_locked(ua_ctx *c, ua_batch *b, int op, int n_runs,
                 return UA_ERR_NOMEM;
             }
         }
-        for (int it = 0; it < n_runs; it++) {
-            bool rec = (it == n_runs - 1);
-            if ((rc = batch_tiles_seq(c, b, kop, stride, rec))) return rc;
-            if ((rc = batch_tail_dispatch(c, b, kop, stride, rec))) return rc;
+        /* overlapped serving shape: the aux tail (scan/compact/pair_out) of
+         * run i executes on the tail stream against buffer set i&1 while
+         * run i+1's tile kernel fills the other set — the ~80 us tail hides
+         * under the ~720 us tile kernel.  Union keeps the serial path (its
+         * WRITE pass is a second tile kernel with live stats events). */
+        bool ovl = (kop != OP_UNION) && n_runs > 1 && c->stream_tail &&
+                   !getenv("UA_NO_OVERLAP");
+        if (ovl && !b->mem1) {
+            size_t o_cnt1 = 0;
+            size_t o_off1 = align16(o_cnt1 + (T + 1) * sizeof(u32));
+            size_t o_part1 = align16(o_off1 + (T + 1) * sizeof(u64));
+            size_t tot1 = align16(o_part1 + (b->nchunks + 1) * sizeof(u64));
+            hipError_t e = hipMalloc(&b->mem1, tot1);
+            if (e == hipSuccess)
+                e = hipMalloc((void **)&b->d_stage1,
+                              (T ? T : 1) * UA_TILE * sizeof(u64));
+            if (e != hipSuccess) {
+                if (b->mem1) { (void)hipFree(b->mem1); b->mem1 = nullptr; }
+                ovl = false; /* fall back to the serial path */
+            } else {
+                u8 *m1 = (u8 *)b->mem1;
+                b->d_tcnt1 = (u32 *)(m1 + o_cnt1);
+                b->d_toff1 = (u64 *)(m1 + o_off1);
+                b->d_part1 = (u64 *)(m1 + o_part1);
+                HIP_TRY(hipMemsetAsync(b->d_tcnt1 + T, 0, sizeof(u32), c->stream));
+            }
+        } else if (ovl && !b->d_stage1) {
+            ovl = false;
         }
-        HIP_TRY(hipStreamSynchronize(c->stream));
+        if (ovl) {
+            for (int it = 0; it < n_runs; it++) {
+                bool rec = (it == n_runs - 1);
+                int s = it & 1;
+                /* set s's tail from run it-2 must be done before refilling */
+                HIP_TRY(hipStreamWaitEvent(c->stream, c->oev[2 + s], 0));
+                if ((rc = batch_tiles_seq(c, b, kop, stride, rec, s))) return rc;
+                HIP_TRY(hipEventRecord(c->oev[s], c->stream));
+                HIP_TRY(hipStreamWaitEvent(c->stream_tail, c->oev[s], 0));
+                if ((rc = batch_tail_dispatch(c, b, kop, stride, rec, s,
+                                              c->stream_tail)))
+                    return rc;
+                HIP_TRY(hipEventRecord(c->oev[2 + s], c->stream_tail));
+            }
+            HIP_TRY(hipStreamSynchronize(c->stream_tail));
+            HIP_TRY(hipStreamSynchronize(c->stream));
+        } else {
+            for (int it = 0; it < n_runs; it++) {
+                bool rec = (it == n_runs - 1);
+                if ((rc = batch_tiles_seq(c, b, kop, stride, rec, 0))) return rc;
+                if ((rc = batch_tail_dispatch(c, b, kop, stride, rec, 0, c->stream)))
+                    return rc;
+            }
+            HIP_TRY(hipStreamSynchronize(c->stream));
+        }
         HIP_TRY(hipGetLastError());
         memcpy(out_lens, b->h_pout, (size_t)b->n_pairs * sizeof(u64));
     }
