@@ -3167,6 +3167,7 @@ static int batch_tail_seq(ua_ctx *c, ua_batch *b, int kop, u64 stride,
 static int batch_tail_dispatch(ua_ctx *c, ua_batch *b, int kop, u64 stride,
                                bool record_events, int set, hipStream_t st) {
 #if UA_GRAPH
+    if (getenv("UA_NO_TAILGRAPH")) b->no_graph = true;
     if (kop != OP_UNION) { /* union's tail holds an event-timed kernel */
         if (!b->gexec[kop][set] && !b->no_graph) {
             /* capture the tail once and VALIDATE with a launch+sync; any
@@ -3295,13 +3296,19 @@ static int batch_run_locked(ua_ctx *c, ua_batch *b, int op, int n_runs,
                 return UA_ERR_NOMEM;
             }
         }
-        /* overlapped serving shape: the aux tail (scan/compact/pair_out) of
-         * run i executes on the tail stream against buffer set i&1 while
-         * run i+1's tile kernel fills the other set — the ~80 us tail hides
-         * under the ~720 us tile kernel.  Union keeps the serial path (its
-         * WRITE pass is a second tile kernel with live stats events). */
+        /* overlapped serving shape (UA_OVERLAP=1, measured NEUTRAL and
+         * default-off): the aux tail (scan/compact/pair_out) of run i
+         * executes on the tail stream against buffer set i&1 while run
+         * i+1's tile kernel fills the other set.  Parity-green, but
+         * co-scheduling contention stretches the tile kernel (727->766 us)
+         * and k_compact (64->90 us) by about what the overlap hides —
+         * whole-step 0.813 vs 0.802 ms serial, with one box showing a
+         * pathological 1.65 ms at long run counts.  Union always serial
+         * (its WRITE pass is a second tile kernel with live stats
+         * events). */
+        const char *ovl_env = getenv("UA_OVERLAP");
         bool ovl = (kop != OP_UNION) && n_runs > 1 && c->stream_tail &&
-                   !getenv("UA_NO_OVERLAP");
+                   ovl_env && ovl_env[0] == '1';
         if (ovl && !b->mem1) {
             size_t o_cnt1 = 0;
             size_t o_off1 = align16(o_cnt1 + (T + 1) * sizeof(u32));
