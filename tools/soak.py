@@ -32,11 +32,13 @@ def main():
 
     eng = algo.Engine(0)
     batch = eng.make_batch(us, vs, outs)
+    print("batch created", flush=True)
     ref = batch.run(algo.OP_INTERSECT)
+    print("single run ok", flush=True)
 
     lens = batch.run_n(algo.OP_INTERSECT, steps)
     assert lens == ref, "run_n soak diverged from single run"
-    print(f"soak ok: {steps} pipelined runs, lens stable ({sum(ref)} total)")
+    print(f"soak ok: {steps} pipelined runs, lens stable ({sum(ref)} total)", flush=True)
 
     base = free_mb()
     for i in range(100):
