@@ -28,7 +28,10 @@ def main():
         up, vp, _ = synth.offset_pair(u0, v0, common0, p)
         us.append(torch.from_numpy(up.view(np.int64)).cuda())
         vs.append(torch.from_numpy(vp.view(np.int64)).cuda())
-        outs.append(torch.empty(1_000_000, dtype=torch.int64, device="cuda"))
+        # n+m capacity: phase 2 runs OP_MERGE too (union outputs up to n+m;
+        # the 1M-cap first version of this script was itself the OOB bug the
+        # first soak run "found")
+        outs.append(torch.empty(2_000_000, dtype=torch.int64, device="cuda"))
 
     eng = algo.Engine(0)
     batch = eng.make_batch(us, vs, outs)
