@@ -126,7 +126,10 @@ class Engine:
             pairs[i].out = outs[i].data_ptr()
         h = C.c_void_p()
         check(lib().ua_batch_create(self._ctx, pairs, np_, C.byref(h)))
-        return Batch(self, h, np_, (us, vs, outs))
+        b = Batch(self, h, np_, (us, vs, outs))
+        b._caps = [(us[i].numel(), vs[i].numel(), outs[i].numel())
+                   for i in range(np_)]
+        return b
 
     def intersect_sorted(self, lists):
         """algo.IntersectSorted (uidlist.go:297): k-way fold, smallest first."""
@@ -407,7 +410,23 @@ class Batch:
         self.n_pairs = n_pairs
         self._keep = keepalive  # input/output tensors must outlive the batch
 
+    def _check_caps(self, op):
+        """Per-pair output capacity contract (uidalgo.h: intersect needs
+        min(n,m), difference n, union n+m).  The C ABI sees raw pointers;
+        this layer knows the tensor sizes, so an undersized output becomes a
+        clean error instead of OOB device writes."""
+        caps = getattr(self, "_caps", None)
+        if caps is None:
+            return
+        for i, (n, m, o) in enumerate(caps):
+            need = min(n, m) if op == OP_INTERSECT else (
+                n if op == OP_DIFFERENCE else n + m)
+            if o < need:
+                raise ValueError(
+                    f"pair {i}: out capacity {o} < required {need} for op {op}")
+
     def run(self, op=OP_INTERSECT):
+        self._check_caps(op)
         lens = (C.c_uint64 * self.n_pairs)()
         check(lib().ua_batch_run(self._eng._ctx, self._h, op, lens))
         return [int(lens[i]) for i in range(self.n_pairs)]
@@ -415,6 +434,7 @@ class Batch:
     def run_n(self, op, n_runs):
         """n_runs passes enqueued back-to-back, ONE sync at the end (the
         repeated-query serving shape — no host round-trip between runs)."""
+        self._check_caps(op)
         lens = (C.c_uint64 * self.n_pairs)()
         check(lib().ua_batch_run_n(self._eng._ctx, self._h, op, n_runs, lens))
         return [int(lens[i]) for i in range(self.n_pairs)]
