@@ -676,3 +676,22 @@ def test_batch_run_n_matches_run(eng):
         for i in range(3):
             assert np.array_equal(to_np(outs[i][:lens5[i]]), snap1[i])
     b.close()
+
+
+def test_batch_out_capacity_guard(eng):
+    """Undersized outputs raise a clean ValueError BEFORE any launch
+    (uidalgo.h capacity contract: intersect min(n,m), diff n, union n+m).
+    Found by the r02 soak: a union run into min(n,m)-sized outs is OOB
+    device writes the raw-pointer C ABI cannot catch."""
+    rng = np.random.default_rng(SEED + 99)
+    u = to_dev(np.sort(rng.choice(100_000, size=5000, replace=False)).astype(np.uint64))
+    v = to_dev(np.sort(rng.choice(100_000, size=4000, replace=False)).astype(np.uint64))
+    small = torch.empty(4000, dtype=torch.int64, device="cuda:0")  # min(n,m)
+    b = eng.make_batch([u], [v], [small])
+    from dgraph_amd.algo import OP_INTERSECT, OP_MERGE, OP_DIFFERENCE
+    b.run(OP_INTERSECT)  # fits
+    with pytest.raises(ValueError):
+        b.run(OP_MERGE)  # needs n+m
+    with pytest.raises(ValueError):
+        b.run_n(OP_DIFFERENCE, 2)  # needs n = 5000
+    b.close()
