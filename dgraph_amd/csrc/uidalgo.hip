@@ -836,6 +836,11 @@ __global__ __launch_bounds__(UA_TBLOCK) void k_tiles(
     bool has_ab = (a0 > 0);
     bool has_bb = (b0 > 0);
     bool has_bn = ((u64)b1 < d.m);
+#if UA_ABLATE == 5 /* meta + dispatch floor: no fill, no walk, no scan */
+    if (tid == 0) tile_cnt[t] = (u32)((a0 ^ (u32)d.m) & 1u);
+    (void)has_ab; (void)has_bb; (void)has_bn;
+    return;
+#endif
 #if UA_ABLATE != 2
 #if UA_PAD32
     {
