@@ -1962,7 +1962,14 @@ __global__ __launch_bounds__(UA_BLOCK) void k_aisect(
         if (nb > 0) {
             u64 b = d.v[ib + ((lane < nb) ? lane : nb - 1)];
             int pos = d_wave_lb(b, nb, a);
-            eq = (pos < nb) && (__shfl(b, pos) == a);
+            /* UNCONDITIONAL shuffle: under the && short-circuit, lanes with
+             * pos >= nb go inactive, and a bpermute SOURCE lane that is
+             * inactive in the instruction contributes garbage — which is
+             * exactly what the read lane (pos) may be.  Clamp and shuffle
+             * with full exec, then mask. */
+            int posc = pos < nb ? pos : 0;
+            u64 bv = __shfl(b, posc);
+            eq = (pos < nb) && (bv == a);
             b_hi = __shfl(b, nb - 1);
             /* B consumes elements <= this window's A max: later A values
              * are strictly larger (duplicate-free contract), so those b
