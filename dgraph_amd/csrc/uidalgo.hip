@@ -1951,22 +1951,28 @@ __global__ __launch_bounds__(UA_BLOCK) void k_aisect(
     u64 *dst = staging + t * UA_AT;
     u32 running = 0;
 
+    /* register windows, loaded ONCE per element and carried across
+     * iterations by cross-lane rotation (re-loading the full window every
+     * iteration would re-read every element ~2x on balanced inputs and
+     * double the HBM traffic).  Invariants: a[lane] = A[min(ia+lane,
+     * ia_end-1)]; b[lane] = B[min(ib+lane, m-1)] (when m > 0). */
+    u64 a = d.u[(ia + (u64)lane < ia_end) ? ia + (u64)lane : ia_end - 1];
+    u64 b = (m > 0) ? d.v[(ib + (u64)lane < m) ? ib + (u64)lane : m - 1] : 0;
+
     while (ia < ia_end) {
         int na = (int)((ia_end - ia < 64) ? (ia_end - ia) : 64);
         int nb = (int)((m - ib < 64) ? (m - ib) : 64);
         bool valid_a = lane < na;
-        u64 a = d.u[ia + (valid_a ? lane : na - 1)];
         bool b_exh = (ib + (u64)nb >= m);
         bool eq = false;
         u64 b_hi = 0;
+        int cB = 0;
         if (nb > 0) {
-            u64 b = d.v[ib + ((lane < nb) ? lane : nb - 1)];
             int pos = d_wave_lb(b, nb, a);
-            /* UNCONDITIONAL shuffle: under the && short-circuit, lanes with
-             * pos >= nb go inactive, and a bpermute SOURCE lane that is
-             * inactive in the instruction contributes garbage — which is
-             * exactly what the read lane (pos) may be.  Clamp and shuffle
-             * with full exec, then mask. */
+            /* UNCONDITIONAL shuffle: under a && short-circuit, lanes with
+             * pos >= nb would go inactive, and a bpermute SOURCE lane that
+             * is inactive in the instruction contributes garbage — which is
+             * exactly what the read lane (pos) may be. */
             int posc = pos < nb ? pos : 0;
             u64 bv = __shfl(b, posc);
             eq = (pos < nb) && (bv == a);
@@ -1976,7 +1982,7 @@ __global__ __launch_bounds__(UA_BLOCK) void k_aisect(
              * can never match again */
             u64 a_hi = __shfl(a, na - 1);
             u64 bm = __ballot((lane < nb) && (b <= a_hi));
-            ib += (u64)__popcll(bm);
+            cB = __popcll(bm);
         }
         bool consume = valid_a && (b_exh || a <= b_hi);
         bool emit = (OP == OP_INTERSECT) ? (consume && eq) : (consume && !eq);
@@ -1987,8 +1993,24 @@ __global__ __launch_bounds__(UA_BLOCK) void k_aisect(
         }
         running += (u32)__popcll(em);
         u64 cm = __ballot(consume);
-        ia += (u64)__popcll(cm);
-        if (nb == 0 && __popcll(cm) == 0) break; /* no progress possible */
+        int cA = __popcll(cm);
+        if (nb == 0 && cA == 0) break; /* no progress possible */
+        ia += (u64)cA;
+        ib += (u64)cB;
+        /* rotate windows down by the consumed count; only the freed lanes
+         * load fresh elements (clamped, keeping the invariant) */
+        if (cA > 0) {
+            u64 ar = __shfl(a, (lane + cA) & 63);
+            a = (lane >= 64 - cA)
+                    ? d.u[(ia + (u64)lane < ia_end) ? ia + (u64)lane : ia_end - 1]
+                    : ar;
+        }
+        if (cB > 0) {
+            u64 br = __shfl(b, (lane + cB) & 63);
+            b = (lane >= 64 - cB)
+                    ? d.v[(ib + (u64)lane < m) ? ib + (u64)lane : m - 1]
+                    : br;
+        }
     }
     if (lane == 0) tile_cnt[t] = running;
 }
