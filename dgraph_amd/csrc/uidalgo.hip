@@ -1951,28 +1951,48 @@ __global__ __launch_bounds__(UA_BLOCK) void k_aisect(
     u64 *dst = staging + t * UA_AT;
     u32 running = 0;
 
-    /* register windows, loaded ONCE per element and carried across
-     * iterations by cross-lane rotation (re-loading the full window every
-     * iteration would re-read every element ~2x on balanced inputs and
-     * double the HBM traffic).  Invariants: a[lane] = A[min(ia+lane,
-     * ia_end-1)]; b[lane] = B[min(ib+lane, m-1)] (when m > 0). */
-    u64 a = d.u[(ia + (u64)lane < ia_end) ? ia + (u64)lane : ia_end - 1];
-    u64 b = (m > 0) ? d.v[(ib + (u64)lane < m) ? ib + (u64)lane : m - 1] : 0;
+    /* 3-deep register FIFO per side: regs xk[lane] = X[clamp(base+64k+lane)]
+     * for k=0..2; the current 64-wide window [w, w+64) satisfies
+     * w - base < 64, so a window value is a 2-shuffle select over x0/x1 and
+     * x2 is PURE PREFETCH (its load latency hides behind ~a window of
+     * compute before a shift consumes it).  Reloading per window put the
+     * dependent load latency on the serial chain — measured 0.92 ms; the
+     * FIFO takes the loads off the chain. */
+    u64 baseA = ia, baseB = ib;
+    u64 a0r = d.u[(baseA + (u64)lane < ia_end) ? baseA + (u64)lane : ia_end - 1];
+    u64 a1r = d.u[(baseA + 64 + (u64)lane < ia_end) ? baseA + 64 + (u64)lane
+                                                    : ia_end - 1];
+    u64 a2r = d.u[(baseA + 128 + (u64)lane < ia_end) ? baseA + 128 + (u64)lane
+                                                     : ia_end - 1];
+    u64 b0r = 0, b1r = 0, b2r = 0;
+    if (m > 0) {
+        b0r = d.v[(baseB + (u64)lane < m) ? baseB + (u64)lane : m - 1];
+        b1r = d.v[(baseB + 64 + (u64)lane < m) ? baseB + 64 + (u64)lane : m - 1];
+        b2r = d.v[(baseB + 128 + (u64)lane < m) ? baseB + 128 + (u64)lane : m - 1];
+    }
 
     while (ia < ia_end) {
         int na = (int)((ia_end - ia < 64) ? (ia_end - ia) : 64);
         int nb = (int)((m - ib < 64) ? (m - ib) : 64);
         bool valid_a = lane < na;
+        /* materialize the windows: off in [0, 127), never reaches x2 */
+        int offA = (int)(ia - baseA) + lane;
+        u64 aw0 = __shfl(a0r, offA & 63);
+        u64 aw1 = __shfl(a1r, offA & 63);
+        u64 a = (offA < 64) ? aw0 : aw1;
         bool b_exh = (ib + (u64)nb >= m);
         bool eq = false;
         u64 b_hi = 0;
         int cB = 0;
         if (nb > 0) {
+            int offB = (int)(ib - baseB) + lane;
+            u64 bw0 = __shfl(b0r, offB & 63);
+            u64 bw1 = __shfl(b1r, offB & 63);
+            u64 b = (offB < 64) ? bw0 : bw1;
             int pos = d_wave_lb(b, nb, a);
             /* UNCONDITIONAL shuffle: under a && short-circuit, lanes with
              * pos >= nb would go inactive, and a bpermute SOURCE lane that
-             * is inactive in the instruction contributes garbage — which is
-             * exactly what the read lane (pos) may be. */
+             * is inactive in the instruction contributes garbage. */
             int posc = pos < nb ? pos : 0;
             u64 bv = __shfl(b, posc);
             eq = (pos < nb) && (bv == a);
@@ -1997,19 +2017,19 @@ __global__ __launch_bounds__(UA_BLOCK) void k_aisect(
         if (nb == 0 && cA == 0) break; /* no progress possible */
         ia += (u64)cA;
         ib += (u64)cB;
-        /* rotate windows down by the consumed count; only the freed lanes
-         * load fresh elements (clamped, keeping the invariant) */
-        if (cA > 0) {
-            u64 ar = __shfl(a, (lane + cA) & 63);
-            a = (lane >= 64 - cA)
-                    ? d.u[(ia + (u64)lane < ia_end) ? ia + (u64)lane : ia_end - 1]
-                    : ar;
+        if (ia - baseA >= 64) {
+            baseA += 64;
+            a0r = a1r;
+            a1r = a2r;
+            a2r = d.u[(baseA + 128 + (u64)lane < ia_end) ? baseA + 128 + (u64)lane
+                                                         : ia_end - 1];
         }
-        if (cB > 0) {
-            u64 br = __shfl(b, (lane + cB) & 63);
-            b = (lane >= 64 - cB)
-                    ? d.v[(ib + (u64)lane < m) ? ib + (u64)lane : m - 1]
-                    : br;
+        if (m > 0 && ib - baseB >= 64) {
+            baseB += 64;
+            b0r = b1r;
+            b1r = b2r;
+            b2r = d.v[(baseB + 128 + (u64)lane < m) ? baseB + 128 + (u64)lane
+                                                    : m - 1];
         }
     }
     if (lane == 0) tile_cnt[t] = running;
