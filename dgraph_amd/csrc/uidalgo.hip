@@ -782,7 +782,12 @@ __global__ __launch_bounds__(UA_TBLOCK) void k_tiles(
     const UaDesc *__restrict__ descs, const u32 *__restrict__ tile_pair,
     const u32 *__restrict__ tile_a0, u64 total_tiles,
     u64 *__restrict__ staging, u64 stage_stride, u32 *__restrict__ tile_cnt,
-    const u64 *__restrict__ offs, const u64 *__restrict__ partials /* MODE_WRITE only */) {
+    const u64 *__restrict__ offs, const u64 *__restrict__ partials /* MODE_WRITE only */,
+    const unsigned short *__restrict__ isplit_in /* cached per-thread merge-path
+        splits (prepared batch: inputs immutable, so splits are run-invariant
+        like the tile partition); null = compute */,
+    unsigned short *__restrict__ isplit_out /* store splits here on the first
+        run; null = don't */) {
     __shared__ __align__(16) u64 smem[UA_SMEMN];
     __shared__ u32 scan[UA_TBLOCK / 64]; /* per-wave totals for d_block_scan */
     __shared__ u64 s_abefore;
@@ -803,6 +808,10 @@ __global__ __launch_bounds__(UA_TBLOCK) void k_tiles(
     u32 a1 = (t + 1 < total_tiles && tile_pair[t + 1] == p) ? tile_a0[t + 1] : (u32)d.n;
     u32 b0 = (u32)(d0 - a0), b1 = (u32)(d1 - a1);
     int alen = (int)(a1 - a0), blen = (int)(b1 - b0);
+    /* cached split: issued HERE so the 2B load's latency hides under the
+     * fill drain; consumed only after the barrier */
+    int i0_cached = -1;
+    if (isplit_in) i0_cached = (int)isplit_in[t * UA_TBLOCK + tid];
 
 #ifndef UA_GLDS
 #define UA_GLDS 1 /* global_load_lds (LDS-DMA) fill: +13% vs the 16B reg-staged fill (4.03 vs 3.58 TB/s on cfg2); 0 = reg-staged */
@@ -950,7 +959,9 @@ __global__ __launch_bounds__(UA_TBLOCK) void k_tiles(
         int s1 = s0 + UA_WPT;
         if (s0 > tilelen) s0 = tilelen;
         if (s1 > tilelen) s1 = tilelen;
-        int i0 = d_merge_path_px(smem, aoff, alen, boff, blen, s0);
+        int i0 = (i0_cached >= 0) ? i0_cached
+                                  : d_merge_path_px(smem, aoff, alen, boff, blen, s0);
+        if (isplit_out) isplit_out[t * UA_TBLOCK + tid] = (unsigned short)i0;
         u64 a_before = s_abefore;
         u64 b_before = (OP == OP_UNION) ? s_bbefore : 0;
 #ifndef UA_WALK2X
@@ -2649,7 +2660,9 @@ __global__ __launch_bounds__(UA_BLOCK) void k_copy_len(
 template <int OP, int MODE>
 static void launch_tiles(ua_ctx *c, const UaDesc *descs, const u32 *tpair,
                          const u32 *ta0, u64 T, u64 *stage, u64 stride, u32 *tcnt,
-                         const u64 *offs, const u64 *part);
+                         const u64 *offs, const u64 *part,
+                         const unsigned short *isin = nullptr,
+                         unsigned short *isout = nullptr);
 
 /* ==================== host shim ==================== */
 
@@ -2902,7 +2915,8 @@ static u32 rp_grid(const void *kfn, u64 T) {
 template <int OP, int MODE>
 static void launch_tiles(ua_ctx *c, const UaDesc *descs, const u32 *tpair,
                          const u32 *ta0, u64 T, u64 *stage, u64 stride, u32 *tcnt,
-                         const u64 *offs, const u64 *part) {
+                         const u64 *offs, const u64 *part,
+                         const unsigned short *isin, unsigned short *isout) {
     if constexpr (MODE != MODE_LOOKBACK) {
         if (pp_enabled()) {
             u32 G = rp_grid((const void *)k_tiles_pp<OP, MODE>, T);
@@ -2937,7 +2951,8 @@ static void launch_tiles(ua_ctx *c, const UaDesc *descs, const u32 *tpair,
     }
 #endif
     hipLaunchKernelGGL((k_tiles<OP, MODE>), dim3((u32)T), dim3(UA_TBLOCK), 0, c->stream,
-                       descs, tpair, ta0, T, stage, stride, tcnt, offs, part);
+                       descs, tpair, ta0, T, stage, stride, tcnt, offs, part,
+                       isin, isout);
 }
 
 /* Split flat scan: offs_dev gets chunk-local exclusive offsets, WS_PARTIAL
@@ -3164,6 +3179,12 @@ struct ua_batch {
     u64 *d_pout = nullptr;
     u64 *d_lbf = nullptr;    /* lookback flag array [total_tiles] */
     u32 lb_gen = 0;
+    /* cached per-thread merge-path splits (valid because the prepared
+     * batch's inputs are immutable — same contract as the cached tile
+     * partition): u16 per (tile, thread); written by the first staged /
+     * lookback run, read by every later one */
+    unsigned short *d_isplit = nullptr;
+    bool i0_ready = false;
     /* A-indexed layout for the wave-register intersect (k_aisect) */
     u64 total_atiles = 0;
     u64 nchunks_a = 0;
@@ -3294,6 +3315,7 @@ extern "C" void ua_batch_destroy(ua_ctx *c, ua_batch *b) {
     if (b->mem1) (void)hipFree(b->mem1);
     if (b->d_stage) (void)hipFree(b->d_stage);
     if (b->d_stage1) (void)hipFree(b->d_stage1);
+    if (b->d_isplit) (void)hipFree(b->d_isplit);
     delete b;
 }
 
@@ -3317,11 +3339,38 @@ static int aisect_enabled() {
     return v;
 }
 
+/* lazy split-cache: returns (isin, isout) for this run and flips i0_ready */
+static void batch_isplit(ua_batch *b, const unsigned short **isin,
+                         unsigned short **isout) {
+    *isin = nullptr;
+    *isout = nullptr;
+    /* the experimental kernels (UA_PP / UA_RPIPE envs) neither store nor
+     * read the cache — engaging i0_ready with them active would hand a
+     * later default-kernel run an unwritten buffer */
+    if (pp_enabled() || rp_enabled()) return;
+    if (!b->d_isplit) {
+        size_t bytes = (size_t)b->total_tiles * UA_TBLOCK * sizeof(unsigned short);
+        if (bytes == 0 || hipMalloc((void **)&b->d_isplit, bytes) != hipSuccess) {
+            b->d_isplit = nullptr; /* cacheless fallback */
+            (void)hipGetLastError();
+            return;
+        }
+    }
+    if (b->i0_ready) *isin = b->d_isplit;
+    else {
+        *isout = b->d_isplit;
+        b->i0_ready = true; /* ordered: runs are stream-serialized */
+    }
+}
+
 static int batch_tiles_seq(ua_ctx *c, ua_batch *b, int kop, u64 stride,
                            bool record_events, int set) {
     u64 T = b->total_tiles;
     u64 *stage = set ? b->d_stage1 : b->d_stage;
     u32 *tcnt = set ? b->d_tcnt1 : b->d_tcnt;
+    const unsigned short *isin;
+    unsigned short *isout;
+    batch_isplit(b, &isin, &isout);
     if ((kop == OP_INTERSECT || kop == OP_DIFF) && aisect_enabled() &&
         b->total_atiles > 0) {
         u64 Ta = b->total_atiles;
@@ -3343,14 +3392,15 @@ static int batch_tiles_seq(ua_ctx *c, ua_batch *b, int kop, u64 stride,
     if (kop == OP_INTERSECT) {
         launch_tiles<OP_INTERSECT, MODE_STAGE>(c, b->d_descs, b->d_tpair, b->d_ta0,
                                                T, stage, stride, tcnt,
-                                               nullptr, nullptr);
+                                               nullptr, nullptr, isin, isout);
     } else if (kop == OP_DIFF) {
         launch_tiles<OP_DIFF, MODE_STAGE>(c, b->d_descs, b->d_tpair, b->d_ta0, T,
                                           stage, stride, tcnt, nullptr,
-                                          nullptr);
+                                          nullptr, isin, isout);
     } else {
         launch_tiles<OP_UNION, MODE_COUNT>(c, b->d_descs, b->d_tpair, b->d_ta0, T,
-                                           nullptr, 0, b->d_tcnt, nullptr, nullptr);
+                                           nullptr, 0, b->d_tcnt, nullptr, nullptr,
+                                           isin, isout);
     }
     if (record_events) HIP_TRY(hipEventRecord(c->ev[1], c->stream));
     return UA_OK;
@@ -3379,10 +3429,13 @@ static int batch_tail_seq(ua_ctx *c, ua_batch *b, int kop, u64 stride,
     hipLaunchKernelGGL(k_scan2, dim3(1), dim3(UA_BLOCK), 0, st, part,
                        nch);
     if (kop == OP_UNION) {
+        /* the count pass of this same run already stored the splits */
+        const unsigned short *isw =
+            (b->i0_ready && b->d_isplit) ? b->d_isplit : nullptr;
         if (record_events) HIP_TRY(hipEventRecord(c->ev[2], st));
         launch_tiles<OP_UNION, MODE_WRITE>(c, b->d_descs, b->d_tpair, b->d_ta0, T,
                                            nullptr, 0, b->d_tcnt, b->d_toff,
-                                           b->d_part);
+                                           b->d_part, isw, nullptr);
         if (record_events) HIP_TRY(hipEventRecord(c->ev[3], st));
     } else {
         hipLaunchKernelGGL(k_compact, dim3((u32)((T + 15) / 16)), dim3(UA_BLOCK), 0,
@@ -3492,19 +3545,25 @@ static int batch_run_locked(ua_ctx *c, ua_batch *b, int op, int n_runs,
             }
             b->lb_gen += 1;
             u64 gen = b->lb_gen;
+            const unsigned short *isin;
+            unsigned short *isout;
+            batch_isplit(b, &isin, &isout);
             if (rec) HIP_TRY(hipEventRecord(c->ev[0], c->stream));
             if (kop == OP_INTERSECT) {
                 launch_tiles<OP_INTERSECT, MODE_LOOKBACK>(c, b->d_descs, b->d_tpair,
                                                           b->d_ta0, T, b->d_lbf, gen,
-                                                          nullptr, b->d_pout, nullptr);
+                                                          nullptr, b->d_pout, nullptr,
+                                                          isin, isout);
             } else if (kop == OP_DIFF) {
                 launch_tiles<OP_DIFF, MODE_LOOKBACK>(c, b->d_descs, b->d_tpair,
                                                      b->d_ta0, T, b->d_lbf, gen,
-                                                     nullptr, b->d_pout, nullptr);
+                                                     nullptr, b->d_pout, nullptr,
+                                                     isin, isout);
             } else {
                 launch_tiles<OP_UNION, MODE_LOOKBACK>(c, b->d_descs, b->d_tpair,
                                                       b->d_ta0, T, b->d_lbf, gen,
-                                                      nullptr, b->d_pout, nullptr);
+                                                      nullptr, b->d_pout, nullptr,
+                                                      isin, isout);
             }
             if (rec) HIP_TRY(hipEventRecord(c->ev[1], c->stream));
         }
