@@ -28,13 +28,15 @@ def main():
         us.append(torch.from_numpy(up.view(np.int64)).cuda())
         vs.append(torch.from_numpy(vp.view(np.int64)).cuda())
     outs = [torch.empty(1_000_000, dtype=torch.int64, device="cuda") for _ in range(pairs)]
+    # prepared batch: mirrors bench.py (partition + split cache warm after
+    # the warmup runs)
+    batch = eng.make_batch(us, vs, outs)
     for _ in range(5):
-        eng.intersect_pairs(us, vs, outs)
+        batch.run(algo.OP_INTERSECT)
     torch.cuda.synchronize()
     eng.stats_reset()
     t0 = time.perf_counter()
-    for _ in range(steps):
-        eng.intersect_pairs(us, vs, outs)
+    batch.run_n(algo.OP_INTERSECT, steps)
     torch.cuda.synchronize()
     el = time.perf_counter() - t0
     st = eng.stats()
