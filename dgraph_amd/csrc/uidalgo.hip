@@ -1774,7 +1774,9 @@ __global__ __launch_bounds__(UA_TBLOCK, 8) void k_tiles_pp(
     const UaDesc *__restrict__ descs, const u32 *__restrict__ tile_pair,
     const u32 *__restrict__ tile_a0, u64 total_tiles,
     u64 *__restrict__ staging, u64 stage_stride, u32 *__restrict__ tile_cnt,
-    const u64 *__restrict__ offs, const u64 *__restrict__ partials) {
+    const u64 *__restrict__ offs, const u64 *__restrict__ partials,
+    const unsigned short *__restrict__ isplit_in,
+    unsigned short *__restrict__ isplit_out) {
     static_assert(UA_TBLOCK == UA_BLOCK, "d_p2_issue strides are 4-wave");
     __shared__ __align__(16) u64 smem[2][UA_TILE + 4];
     __shared__ u32 scanb[UA_TBLOCK / 64];
@@ -1814,7 +1816,9 @@ __global__ __launch_bounds__(UA_TBLOCK, 8) void k_tiles_pp(
         u64 em[UA_WPT];
         u32 flags = 0, amask = 0;
         int cnt = 0, w_i0 = 0;
-        int i0 = d_merge_path_px(buf, f.aoff, alen, f.boff, blen, s0);
+        int i0 = isplit_in ? (int)isplit_in[t * UA_TBLOCK + tid]
+                           : d_merge_path_px(buf, f.aoff, alen, f.boff, blen, s0);
+        if (isplit_out) isplit_out[t * UA_TBLOCK + tid] = (unsigned short)i0;
         if (UA_WALK3 && (OP == OP_INTERSECT || OP == OP_DIFF)) {
             cnt = tile_walk3<OP, UA_WPT>(buf, f.aoff, alen, f.boff, blen, m.has_bn,
                                          s0, s1, i0, flags, amask);
@@ -2922,7 +2926,7 @@ static void launch_tiles(ua_ctx *c, const UaDesc *descs, const u32 *tpair,
             u32 G = rp_grid((const void *)k_tiles_pp<OP, MODE>, T);
             hipLaunchKernelGGL((k_tiles_pp<OP, MODE>), dim3(G), dim3(UA_TBLOCK), 0,
                                c->stream, descs, tpair, ta0, T, stage, stride, tcnt,
-                               offs, part);
+                               offs, part, isin, isout);
             return;
         }
         if (rp_enabled()) {
@@ -3344,10 +3348,10 @@ static void batch_isplit(ua_batch *b, const unsigned short **isin,
                          unsigned short **isout) {
     *isin = nullptr;
     *isout = nullptr;
-    /* the experimental kernels (UA_PP / UA_RPIPE envs) neither store nor
-     * read the cache — engaging i0_ready with them active would hand a
-     * later default-kernel run an unwritten buffer */
-    if (pp_enabled() || rp_enabled()) return;
+    /* UA_RPIPE's kernel neither stores nor reads the cache — engaging
+     * i0_ready with it active would hand a later run an unwritten buffer
+     * (k_tiles and k_tiles_pp both participate) */
+    if (rp_enabled()) return;
     if (!b->d_isplit) {
         size_t bytes = (size_t)b->total_tiles * UA_TBLOCK * sizeof(unsigned short);
         if (bytes == 0 || hipMalloc((void **)&b->d_isplit, bytes) != hipSuccess) {
