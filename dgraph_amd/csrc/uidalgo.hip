@@ -664,6 +664,73 @@ __device__ __forceinline__ int tile_walk3(const u64 *smembase, int aoff, int ale
     return cnt;
 }
 
+/* Dual-chain maskless walk (intersect/diff): two independent 4-step
+ * gather chains per thread (halved serial LDS latency), walk3's
+ * no-em[] bookkeeping.  Viable because the prepared batch caches BOTH
+ * splits (i0 at s0, i0b at smid) — round 1 rejected the dual-chain
+ * em[] walk precisely because its second diagonal search cost more
+ * than the added MLP recovered. */
+template <int OP, int W>
+__device__ __forceinline__ int tile_walk3x(const u64 *smembase, int aoff, int alen,
+                                           int boff, int blen, bool has_bn,
+                                           int s0, int smid, int s1, int i0a,
+                                           int i0b, u32 &flags, u32 &amA, u32 &amB) {
+    static_assert(W <= 32, "masks are u32");
+    constexpr int H = W / 2;
+    int blen_ext = blen + (has_bn ? 1 : 0);
+    int amax = alen > 0 ? alen - 1 : 0;
+    int bmax = blen_ext > 0 ? blen_ext - 1 : 0;
+    int ia = i0a, ja = s0 - i0a;
+    int ib = i0b, jb = smid - i0b;
+    int stepsA = smid - s0, stepsB = s1 - smid;
+    u64 aA = smembase[UA_PX(aoff + (ia < alen ? ia : amax))];
+    u64 bA = smembase[UA_PX(boff + (ja < blen_ext ? ja : bmax))];
+    u64 aB = smembase[UA_PX(aoff + (ib < alen ? ib : amax))];
+    u64 bB = smembase[UA_PX(boff + (jb < blen_ext ? jb : bmax))];
+    int cnt = 0;
+    flags = 0;
+    amA = 0;
+    amB = 0;
+#pragma unroll
+    for (int q = 0; q < H; q++) {
+        if (q < stepsA && (ia < alen || ja < blen)) {
+            bool inA = ia < alen, inB = ja < blen;
+            bool takeA = inA && (!inB || aA <= bA);
+            bool eq = (aA == bA) && (ja < blen_ext);
+            bool emit = (OP == OP_INTERSECT) ? (takeA && eq) : (takeA && !eq);
+            flags |= ((u32)emit) << q;
+            amA |= ((u32)takeA) << q;
+            cnt += emit;
+            int ni = ia + (takeA ? 1 : 0), nj = ja + (takeA ? 0 : 1);
+            int raddr = takeA ? (aoff + (ni < alen ? ni : amax))
+                              : (boff + (nj < blen_ext ? nj : bmax));
+            u64 r = smembase[UA_PX(raddr)];
+            aA = takeA ? r : aA;
+            bA = takeA ? bA : r;
+            ia = ni;
+            ja = nj;
+        }
+        if (q < stepsB && (ib < alen || jb < blen)) {
+            bool inA = ib < alen, inB = jb < blen;
+            bool takeA = inA && (!inB || aB <= bB);
+            bool eq = (aB == bB) && (jb < blen_ext);
+            bool emit = (OP == OP_INTERSECT) ? (takeA && eq) : (takeA && !eq);
+            flags |= ((u32)emit) << (H + q);
+            amB |= ((u32)takeA) << q;
+            cnt += emit;
+            int ni = ib + (takeA ? 1 : 0), nj = jb + (takeA ? 0 : 1);
+            int raddr = takeA ? (aoff + (ni < alen ? ni : amax))
+                              : (boff + (nj < blen_ext ? nj : bmax));
+            u64 r = smembase[UA_PX(raddr)];
+            aB = takeA ? r : aB;
+            bB = takeA ? bB : r;
+            ib = ni;
+            jb = nj;
+        }
+    }
+    return cnt;
+}
+
 /* Dual-chain walk (UA_WALK2X): each thread walks TWO independent
  * half-segments (diagonals s0 and s0+W/2) with interleaved steps — two
  * independent LDS-gather dependency chains per thread to close the
@@ -783,11 +850,11 @@ __global__ __launch_bounds__(UA_TBLOCK) void k_tiles(
     const u32 *__restrict__ tile_a0, u64 total_tiles,
     u64 *__restrict__ staging, u64 stage_stride, u32 *__restrict__ tile_cnt,
     const u64 *__restrict__ offs, const u64 *__restrict__ partials /* MODE_WRITE only */,
-    const unsigned short *__restrict__ isplit_in /* cached per-thread merge-path
-        splits (prepared batch: inputs immutable, so splits are run-invariant
-        like the tile partition); null = compute */,
-    unsigned short *__restrict__ isplit_out /* store splits here on the first
-        run; null = don't */) {
+    const u32 *__restrict__ isplit_in /* cached per-thread merge-path splits,
+        PACKED lo16 = i0 at diagonal s0, hi16 = i0b at smid (prepared batch:
+        inputs immutable, so splits are run-invariant like the tile
+        partition); null = compute */,
+    u32 *__restrict__ isplit_out /* store packed splits on the first run */) {
     __shared__ __align__(16) u64 smem[UA_SMEMN];
     __shared__ u32 scan[UA_TBLOCK / 64]; /* per-wave totals for d_block_scan */
     __shared__ u64 s_abefore;
@@ -808,10 +875,14 @@ __global__ __launch_bounds__(UA_TBLOCK) void k_tiles(
     u32 a1 = (t + 1 < total_tiles && tile_pair[t + 1] == p) ? tile_a0[t + 1] : (u32)d.n;
     u32 b0 = (u32)(d0 - a0), b1 = (u32)(d1 - a1);
     int alen = (int)(a1 - a0), blen = (int)(b1 - b0);
-    /* cached split: issued HERE so the 2B load's latency hides under the
+    /* cached splits: issued HERE so the 4B load's latency hides under the
      * fill drain; consumed only after the barrier */
-    int i0_cached = -1;
-    if (isplit_in) i0_cached = (int)isplit_in[t * UA_TBLOCK + tid];
+    int i0_cached = -1, i0b_cached = -1;
+    if (isplit_in) {
+        u32 w = isplit_in[t * UA_TBLOCK + tid];
+        i0_cached = (int)(w & 0xffffu);
+        i0b_cached = (int)(w >> 16);
+    }
 
 #ifndef UA_GLDS
 #define UA_GLDS 1 /* global_load_lds (LDS-DMA) fill: +13% vs the 16B reg-staged fill (4.03 vs 3.58 TB/s on cfg2); 0 = reg-staged */
@@ -935,15 +1006,15 @@ __global__ __launch_bounds__(UA_TBLOCK) void k_tiles(
     asm volatile("" ::"v"(ablate_x));
     u64 em[UA_WPT];
     u32 flags = 0;
-    u32 amask = 0;
-    int w_i0 = 0;
+    u32 amA = 0, amB = 0;
+    int w_i0 = 0, w_i0b = 0;
     int cnt = 0;
     (void)s_abefore;
 #else
     u64 em[UA_WPT];
     u32 flags;
-    u32 amask = 0; /* walk3 (intersect/diff): takeA bits for value reconstruction */
-    int w_i0 = 0;
+    u32 amA = 0, amB = 0; /* walk3x: per-chain takeA bits for reconstruction */
+    int w_i0 = 0, w_i0b = 0;
     int cnt;
 #ifndef UA_WALK2
 #define UA_WALK2 1 /* 0 = the branchy register-frontier walk */
@@ -959,9 +1030,17 @@ __global__ __launch_bounds__(UA_TBLOCK) void k_tiles(
         int s1 = s0 + UA_WPT;
         if (s0 > tilelen) s0 = tilelen;
         if (s1 > tilelen) s1 = tilelen;
+        int smid = s0 + UA_WPT / 2;
+        if (smid > tilelen) smid = tilelen;
+        if (smid < s0) smid = s0;
+        if (smid > s1) smid = s1;
         int i0 = (i0_cached >= 0) ? i0_cached
                                   : d_merge_path_px(smem, aoff, alen, boff, blen, s0);
-        if (isplit_out) isplit_out[t * UA_TBLOCK + tid] = (unsigned short)i0;
+        int i0b = (i0b_cached >= 0)
+                      ? i0b_cached
+                      : d_merge_path_px(smem, aoff, alen, boff, blen, smid);
+        if (isplit_out)
+            isplit_out[t * UA_TBLOCK + tid] = (u32)i0 | ((u32)i0b << 16);
         u64 a_before = s_abefore;
         u64 b_before = (OP == OP_UNION) ? s_bbefore : 0;
 #ifndef UA_WALK2X
@@ -983,9 +1062,10 @@ __global__ __launch_bounds__(UA_TBLOCK) void k_tiles(
         } else
 #endif
         if (UA_WALK3 && (OP == OP_INTERSECT || OP == OP_DIFF)) {
-            cnt = tile_walk3<OP, UA_WPT>(smem, aoff, alen, boff, blen, has_bn,
-                                         s0, s1, i0, flags, amask);
+            cnt = tile_walk3x<OP, UA_WPT>(smem, aoff, alen, boff, blen, has_bn,
+                                          s0, smid, s1, i0, i0b, flags, amA, amB);
             w_i0 = i0;
+            w_i0b = i0b;
         } else if (MODE == MODE_COUNT) {
             /* count pass needs no values — skip the em[] bookkeeping */
             cnt = tile_walk2c<OP, UA_WPT>(smem, aoff, alen, boff, blen, a_before,
@@ -1060,13 +1140,16 @@ __global__ __launch_bounds__(UA_TBLOCK) void k_tiles(
             u64 room = cap - gbase;
             int lim = (int)((room < (u64)cnt) ? room : (u64)cnt);
             if (UA_WALK3 && (OP == OP_INTERSECT || OP == OP_DIFF)) {
-                /* reconstruct emitted A-values from the takeA mask */
+                /* reconstruct emitted A-values from the per-chain masks */
+                constexpr int H = UA_WPT / 2;
                 u32 f = flags;
                 int k = 0;
                 while (f && k < lim) {
                     int s = __builtin_ctz(f);
                     f &= f - 1;
-                    int idx = w_i0 + __popc(amask & ((1u << s) - 1));
+                    int idx = (s < H)
+                                  ? w_i0 + __popc(amA & ((1u << s) - 1))
+                                  : w_i0b + __popc(amB & ((1u << (s - H)) - 1));
                     dst[k++] = smem[UA_PX(aoff + idx)];
                 }
             } else {
@@ -1106,13 +1189,16 @@ __global__ __launch_bounds__(UA_TBLOCK) void k_tiles(
     }
     if (cnt > 0) {
         if (UA_WALK3 && (OP == OP_INTERSECT || OP == OP_DIFF)) {
-            /* reconstruct emitted A-values from the takeA mask */
+            /* reconstruct emitted A-values from the per-chain masks */
+            constexpr int H = UA_WPT / 2;
             u32 f = flags;
             u32 k = 0;
             while (f && k < lim) {
                 int s = __builtin_ctz(f);
                 f &= f - 1;
-                int idx = w_i0 + __popc(amask & ((1u << s) - 1));
+                int idx = (s < H)
+                              ? w_i0 + __popc(amA & ((1u << s) - 1))
+                              : w_i0b + __popc(amB & ((1u << (s - H)) - 1));
                 dst[k++] = smem[UA_PX(aoff + idx)];
             }
         } else {
@@ -1775,8 +1861,7 @@ __global__ __launch_bounds__(UA_TBLOCK, 8) void k_tiles_pp(
     const u32 *__restrict__ tile_a0, u64 total_tiles,
     u64 *__restrict__ staging, u64 stage_stride, u32 *__restrict__ tile_cnt,
     const u64 *__restrict__ offs, const u64 *__restrict__ partials,
-    const unsigned short *__restrict__ isplit_in,
-    unsigned short *__restrict__ isplit_out) {
+    const u32 *__restrict__ isplit_in, u32 *__restrict__ isplit_out) {
     static_assert(UA_TBLOCK == UA_BLOCK, "d_p2_issue strides are 4-wave");
     __shared__ __align__(16) u64 smem[2][UA_TILE + 4];
     __shared__ u32 scanb[UA_TBLOCK / 64];
@@ -1816,9 +1901,9 @@ __global__ __launch_bounds__(UA_TBLOCK, 8) void k_tiles_pp(
         u64 em[UA_WPT];
         u32 flags = 0, amask = 0;
         int cnt = 0, w_i0 = 0;
-        int i0 = isplit_in ? (int)isplit_in[t * UA_TBLOCK + tid]
+        int i0 = isplit_in ? (int)(isplit_in[t * UA_TBLOCK + tid] & 0xffffu)
                            : d_merge_path_px(buf, f.aoff, alen, f.boff, blen, s0);
-        if (isplit_out) isplit_out[t * UA_TBLOCK + tid] = (unsigned short)i0;
+        (void)isplit_out; /* pp never participates in the cache */
         if (UA_WALK3 && (OP == OP_INTERSECT || OP == OP_DIFF)) {
             cnt = tile_walk3<OP, UA_WPT>(buf, f.aoff, alen, f.boff, blen, m.has_bn,
                                          s0, s1, i0, flags, amask);
@@ -2665,8 +2750,7 @@ template <int OP, int MODE>
 static void launch_tiles(ua_ctx *c, const UaDesc *descs, const u32 *tpair,
                          const u32 *ta0, u64 T, u64 *stage, u64 stride, u32 *tcnt,
                          const u64 *offs, const u64 *part,
-                         const unsigned short *isin = nullptr,
-                         unsigned short *isout = nullptr);
+                         const u32 *isin = nullptr, u32 *isout = nullptr);
 
 /* ==================== host shim ==================== */
 
@@ -2920,7 +3004,7 @@ template <int OP, int MODE>
 static void launch_tiles(ua_ctx *c, const UaDesc *descs, const u32 *tpair,
                          const u32 *ta0, u64 T, u64 *stage, u64 stride, u32 *tcnt,
                          const u64 *offs, const u64 *part,
-                         const unsigned short *isin, unsigned short *isout) {
+                         const u32 *isin, u32 *isout) {
     if constexpr (MODE != MODE_LOOKBACK) {
         if (pp_enabled()) {
             u32 G = rp_grid((const void *)k_tiles_pp<OP, MODE>, T);
@@ -3187,7 +3271,7 @@ struct ua_batch {
      * batch's inputs are immutable — same contract as the cached tile
      * partition): u16 per (tile, thread); written by the first staged /
      * lookback run, read by every later one */
-    unsigned short *d_isplit = nullptr;
+    u32 *d_isplit = nullptr; /* packed lo16 = i0(s0), hi16 = i0b(smid) */
     bool i0_ready = false;
     /* A-indexed layout for the wave-register intersect (k_aisect) */
     u64 total_atiles = 0;
@@ -3344,16 +3428,15 @@ static int aisect_enabled() {
 }
 
 /* lazy split-cache: returns (isin, isout) for this run and flips i0_ready */
-static void batch_isplit(ua_batch *b, const unsigned short **isin,
-                         unsigned short **isout) {
+static void batch_isplit(ua_batch *b, const u32 **isin, u32 **isout) {
     *isin = nullptr;
     *isout = nullptr;
-    /* UA_RPIPE's kernel neither stores nor reads the cache — engaging
-     * i0_ready with it active would hand a later run an unwritten buffer
-     * (k_tiles and k_tiles_pp both participate) */
-    if (rp_enabled()) return;
+    /* the experimental kernels neither store nor read the PACKED cache —
+     * engaging i0_ready with them active would hand a later default run an
+     * unwritten buffer */
+    if (pp_enabled() || rp_enabled()) return;
     if (!b->d_isplit) {
-        size_t bytes = (size_t)b->total_tiles * UA_TBLOCK * sizeof(unsigned short);
+        size_t bytes = (size_t)b->total_tiles * UA_TBLOCK * sizeof(u32);
         if (bytes == 0 || hipMalloc((void **)&b->d_isplit, bytes) != hipSuccess) {
             b->d_isplit = nullptr; /* cacheless fallback */
             (void)hipGetLastError();
@@ -3372,8 +3455,8 @@ static int batch_tiles_seq(ua_ctx *c, ua_batch *b, int kop, u64 stride,
     u64 T = b->total_tiles;
     u64 *stage = set ? b->d_stage1 : b->d_stage;
     u32 *tcnt = set ? b->d_tcnt1 : b->d_tcnt;
-    const unsigned short *isin;
-    unsigned short *isout;
+    const u32 *isin;
+    u32 *isout;
     batch_isplit(b, &isin, &isout);
     if ((kop == OP_INTERSECT || kop == OP_DIFF) && aisect_enabled() &&
         b->total_atiles > 0) {
@@ -3434,8 +3517,7 @@ static int batch_tail_seq(ua_ctx *c, ua_batch *b, int kop, u64 stride,
                        nch);
     if (kop == OP_UNION) {
         /* the count pass of this same run already stored the splits */
-        const unsigned short *isw =
-            (b->i0_ready && b->d_isplit) ? b->d_isplit : nullptr;
+        const u32 *isw = (b->i0_ready && b->d_isplit) ? b->d_isplit : nullptr;
         if (record_events) HIP_TRY(hipEventRecord(c->ev[2], st));
         launch_tiles<OP_UNION, MODE_WRITE>(c, b->d_descs, b->d_tpair, b->d_ta0, T,
                                            nullptr, 0, b->d_tcnt, b->d_toff,
@@ -3549,8 +3631,8 @@ static int batch_run_locked(ua_ctx *c, ua_batch *b, int op, int n_runs,
             }
             b->lb_gen += 1;
             u64 gen = b->lb_gen;
-            const unsigned short *isin;
-            unsigned short *isout;
+            const u32 *isin;
+            u32 *isout;
             batch_isplit(b, &isin, &isout);
             if (rec) HIP_TRY(hipEventRecord(c->ev[0], c->stream));
             if (kop == OP_INTERSECT) {
