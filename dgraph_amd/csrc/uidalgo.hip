@@ -1062,10 +1062,20 @@ __global__ __launch_bounds__(UA_TBLOCK) void k_tiles(
         } else
 #endif
         if (UA_WALK3 && (OP == OP_INTERSECT || OP == OP_DIFF)) {
+#ifndef UA_DUALWALK
+#define UA_DUALWALK 1 /* 0 = single-chain walk3 (A/B toggle) */
+#endif
+#if UA_DUALWALK
             cnt = tile_walk3x<OP, UA_WPT>(smem, aoff, alen, boff, blen, has_bn,
                                           s0, smid, s1, i0, i0b, flags, amA, amB);
             w_i0 = i0;
             w_i0b = i0b;
+#else
+            cnt = tile_walk3<OP, UA_WPT>(smem, aoff, alen, boff, blen, has_bn,
+                                         s0, s1, i0, flags, amA);
+            w_i0 = i0;
+            w_i0b = 0; /* no chain-B emissions: flags bits >= H unset */
+#endif
         } else if (MODE == MODE_COUNT) {
             /* count pass needs no values — skip the em[] bookkeeping */
             cnt = tile_walk2c<OP, UA_WPT>(smem, aoff, alen, boff, blen, a_before,
