@@ -1071,10 +1071,16 @@ __global__ __launch_bounds__(UA_TBLOCK) void k_tiles(
             w_i0 = i0;
             w_i0b = i0b;
 #else
+            u32 amall;
             cnt = tile_walk3<OP, UA_WPT>(smem, aoff, alen, boff, blen, has_bn,
-                                         s0, s1, i0, flags, amA);
+                                         s0, s1, i0, flags, amall);
+            /* express the single mask in the dual-reconstruction form:
+             * amB = high half, w_i0b = i0 + takeA count of the low half */
+            constexpr u32 HM = (1u << (UA_WPT / 2)) - 1;
+            amA = amall & HM;
+            amB = amall >> (UA_WPT / 2);
             w_i0 = i0;
-            w_i0b = 0; /* no chain-B emissions: flags bits >= H unset */
+            w_i0b = i0 + __popc(amA);
 #endif
         } else if (MODE == MODE_COUNT) {
             /* count pass needs no values — skip the em[] bookkeeping */
