@@ -141,8 +141,11 @@ int ua_intersect_batch_dev(ua_ctx *, const ua_dpair *pairs, int n_pairs,
 
 /* Prepared batch (the repeated-query path): descriptor upload and the
  * merge-path tile partition are done ONCE at create; each run is launches
- * only.  The pairs' device contents must not change between create and the
- * runs (the cached partition depends on them); out capacities must fit the
+ * only, and the FIRST run additionally caches the per-thread merge-path
+ * splits (4 B/tile-thread, device) that every later run loads instead of
+ * re-searching.  The pairs' device contents must not change between
+ * create and the runs (both caches depend on them; violating this gives
+ * unspecified results but stays memory-safe); out capacities must fit the
  * op (intersect >= min(n,m), merge >= n+m, difference >= n). */
 typedef struct ua_batch ua_batch;
 enum { UA_OP_INTERSECT = 0, UA_OP_MERGE = 1, UA_OP_DIFFERENCE = 2 };
