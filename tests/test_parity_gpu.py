@@ -695,3 +695,37 @@ def test_batch_out_capacity_guard(eng):
     with pytest.raises(ValueError):
         b.run_n(OP_DIFFERENCE, 2)  # needs n = 5000
     b.close()
+
+
+def test_extreme_ratio_pairs(eng):
+    """Tile-skew extreme: a 10M-element list against tiny and
+    mid-sized partners in one batch — a 10M list is ~5.9k merge tiles while
+    its 1k partner contributes none of the path, so per-pair tile counts in
+    the grid differ by three orders of magnitude.  Parity vs the oracle for
+    all three ops, through the prepared batch (split cache exercised by the
+    second intersect run)."""
+    import torch
+    rng = np.random.default_rng(SEED + 424)
+    big = synth.gen_sorted_unique(rng, 10_000_000, 40_000_000)
+    shapes = [
+        (synth.gen_sorted_unique(rng, 1_000, 40_000_000), big),
+        (big, synth.gen_sorted_unique(rng, 1_000, 40_000_000)),
+        (synth.gen_sorted_unique(rng, 1, 40_000_000), big),
+        (big, synth.gen_sorted_unique(rng, 300_000, 40_000_000)),
+    ]
+    us = [to_dev(u) for u, _ in shapes]
+    vs = [to_dev(v) for _, v in shapes]
+    outs = [torch.empty(u.numel() + v.numel(), dtype=torch.int64, device="cuda:0")
+            for u, v in zip(us, vs)]
+    batch = eng.make_batch(us, vs, outs)
+    for op, ref in [(algo.OP_INTERSECT, orc.intersect_with),
+                    (algo.OP_MERGE, lambda u, v: orc.merge_sorted([u, v])),
+                    (algo.OP_DIFFERENCE, orc.difference),
+                    (algo.OP_INTERSECT, orc.intersect_with)]:  # cached re-run
+        lens = batch.run(op)
+        for i, (u, v) in enumerate(shapes):
+            want = ref(u, v)
+            assert lens[i] == want.size, (op, i)
+            got = to_np(outs[i][:lens[i]])
+            assert np.array_equal(got, want), (op, i)
+    batch.close()
