@@ -2260,6 +2260,37 @@ __global__ __launch_bounds__(UA_BLOCK) void k_pair_out(const u64 *__restrict__ o
 
 /* one wavefront per tile, 4 tiles per wave, 16 per workgroup: few blocks to
  * dispatch, and tiles with cnt==0 cost one load */
+#define UA_COMPACT_SMALL 32 /* cnt <= SMALL tiles go to the thread-per-tile
+                              * compactor; the wave compactor keeps the rest
+                              * (headline tiles emit ~10, so one thread per
+                              * tile amortizes the 3-load dependent chain
+                              * over 187k parallel threads instead of 47k
+                              * waves) */
+
+/* thread-per-tile compactor for small counts (1..UA_COMPACT_SMALL) */
+__global__ __launch_bounds__(UA_BLOCK) void k_compact_small(
+    const UaDesc *__restrict__ descs, const u32 *__restrict__ tile_pair,
+    const u64 *__restrict__ tb, const u32 *__restrict__ tile_cnt,
+    const u64 *__restrict__ offs, const u64 *__restrict__ partials,
+    const u64 *__restrict__ staging, u64 stage_stride, u64 total_tiles, int op) {
+    u64 t = (u64)blockIdx.x * UA_BLOCK + threadIdx.x;
+    if (t >= total_tiles) return;
+    u32 cnt = tile_cnt[t];
+    if (cnt == 0 || cnt > UA_COMPACT_SMALL) return;
+    u32 p = tile_pair[t];
+    UaDesc d = descs[p];
+    u64 goff = d_off(offs, partials, t) - d_off(offs, partials, tb[p]);
+    u64 cap = (op == OP_INTERSECT) ? (d.n < d.m ? d.n : d.m)
+              : (op == OP_DIFF) ? d.n
+                                : (d.n + d.m);
+    if (goff >= cap) return;
+    u64 room = cap - goff;
+    if ((u64)cnt > room) cnt = (u32)room;
+    u64 *dst = d.out + goff;
+    const u64 *src = staging + t * stage_stride;
+    for (u32 i = 0; i < cnt; i++) dst[i] = src[i];
+}
+
 __global__ __launch_bounds__(UA_BLOCK) void k_compact(
     const UaDesc *__restrict__ descs, const u32 *__restrict__ tile_pair,
     const u64 *__restrict__ tb /* per-pair first tile in the ACTIVE layout:
@@ -2287,7 +2318,7 @@ __global__ __launch_bounds__(UA_BLOCK) void k_compact(
     for (int q = 0; q < 4; q++) {
         u64 t = base + q;
         u32 cnt = cnts4[q];
-        if (cnt == 0) continue;
+        if (cnt <= UA_COMPACT_SMALL) continue; /* k_compact_small's tiles */
         UaDesc d = descs[pair4[q]];
         u64 goff = off4[q] - d_off(offs, partials, tb[pair4[q]]);
         /* pair out-capacity clamp (invalid duplicate/unsorted inputs must
@@ -3214,6 +3245,11 @@ static int run_batch_locked(ua_ctx *c, const ua_dpair *pairs, int n_pairs,
                 hipLaunchKernelGGL(k_compact, dim3((u32)cblk), dim3(UA_BLOCK), 0,
                                    c->stream, d_descs, d_tpair, d_tb, d_tcnt, d_toff,
                                    d_part, d_stage, stage_stride, total_tiles, op);
+                u64 sblk = (total_tiles + UA_BLOCK - 1) / UA_BLOCK;
+                hipLaunchKernelGGL(k_compact_small, dim3((u32)sblk), dim3(UA_BLOCK),
+                                   0, c->stream, d_descs, d_tpair, d_tb, d_tcnt,
+                                   d_toff, d_part, d_stage, stage_stride,
+                                   total_tiles, op);
             }
             u64 poutblk = ((u64)n_pairs + UA_BLOCK - 1) / UA_BLOCK;
             hipLaunchKernelGGL(k_pair_out, dim3((u32)poutblk), dim3(UA_BLOCK), 0,
@@ -3543,6 +3579,9 @@ static int batch_tail_seq(ua_ctx *c, ua_batch *b, int kop, u64 stride,
         hipLaunchKernelGGL(k_compact, dim3((u32)((T + 15) / 16)), dim3(UA_BLOCK), 0,
                            st, b->d_descs, tpair, tbx, tcnt, toff,
                            part, stage, stride_x, T, kop);
+        hipLaunchKernelGGL(k_compact_small, dim3((u32)((T + UA_BLOCK - 1) / UA_BLOCK)),
+                           dim3(UA_BLOCK), 0, st, b->d_descs, tpair, tbx, tcnt,
+                           toff, part, stage, stride_x, T, kop);
     }
     u64 poutblk = ((u64)b->n_pairs + UA_BLOCK - 1) / UA_BLOCK;
     hipLaunchKernelGGL(k_pair_out, dim3((u32)poutblk), dim3(UA_BLOCK), 0, st,
