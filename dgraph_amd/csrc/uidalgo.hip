@@ -74,6 +74,11 @@ typedef uint8_t u8;
 #define UA_SMEMN (UA_TILE + 4)
 #endif
 #define UA_SCAN_CHUNK 2048
+#define UA_STAGE_P 64 /* dense primary staging slots (u64/tile): tiles whose
+                       * total emission fits go here (512 B apart -> DRAM
+                       * row-local compaction reads) instead of the sparse
+                       * stage_stride region (8-16 KB apart), which stays as
+                       * the overflow path for dense-overlap tiles */
 #define UA_PKW 4          /* packed-decode waves (blocks) per workgroup */
 #define UA_MAX_BLOCK_UIDS 256
 #define UA_MAX_DELTAS 1092 /* 64 groups x 17 B, padded */
@@ -1193,10 +1198,16 @@ __global__ __launch_bounds__(UA_TBLOCK) void k_tiles(
     if (MODE == MODE_STAGE) {
         /* clamp to the staging stride: invalid (duplicate/unsorted) inputs
          * can emit more than stride entries per tile; results there are
-         * unspecified, OOB writes are not */
-        u32 c0 = excl < (u32)stage_stride ? (u32)stage_stride - excl : 0;
+         * unspecified, OOB writes are not.  Small totals go to
+         * the dense primary region (offs = its base, see batch_tiles_seq);
+         * stage_stride stays the sparse overflow. */
+        bool prim = (offs != nullptr) && (total <= UA_STAGE_P);
+        u32 cap_t = prim ? (u32)UA_STAGE_P : (u32)stage_stride;
+        u32 c0 = excl < cap_t ? cap_t - excl : 0;
         if (lim > c0) lim = c0;
-        dst = staging + t * stage_stride + excl;
+        dst = prim ? ((u64 *)offs + t * UA_STAGE_P)
+                   : (staging + t * stage_stride + excl);
+        if (prim) dst += excl;
         if (tid == 0)
             tile_cnt[t] = total < (u32)stage_stride ? total : (u32)stage_stride;
     } else { /* MODE_WRITE: (offs, partials) is the split flat scan */
@@ -2272,7 +2283,8 @@ __global__ __launch_bounds__(UA_BLOCK) void k_compact_small(
     const UaDesc *__restrict__ descs, const u32 *__restrict__ tile_pair,
     const u64 *__restrict__ tb, const u32 *__restrict__ tile_cnt,
     const u64 *__restrict__ offs, const u64 *__restrict__ partials,
-    const u64 *__restrict__ staging, u64 stage_stride, u64 total_tiles, int op) {
+    const u64 *__restrict__ staging, u64 stage_stride,
+    const u64 *__restrict__ prim, u64 total_tiles, int op) {
     u64 t = (u64)blockIdx.x * UA_BLOCK + threadIdx.x;
     if (t >= total_tiles) return;
     u32 cnt = tile_cnt[t];
@@ -2287,7 +2299,7 @@ __global__ __launch_bounds__(UA_BLOCK) void k_compact_small(
     u64 room = cap - goff;
     if ((u64)cnt > room) cnt = (u32)room;
     u64 *dst = d.out + goff;
-    const u64 *src = staging + t * stage_stride;
+    const u64 *src = prim ? prim + t * UA_STAGE_P : staging + t * stage_stride;
     for (u32 i = 0; i < cnt; i++) dst[i] = src[i];
 }
 
@@ -2298,7 +2310,7 @@ __global__ __launch_bounds__(UA_BLOCK) void k_compact(
                                   for the A-indexed (k_aisect) layout */,
     const u32 *__restrict__ tile_cnt, const u64 *__restrict__ offs,
     const u64 *__restrict__ partials, const u64 *__restrict__ staging,
-    u64 stage_stride, u64 total_tiles, int op) {
+    u64 stage_stride, const u64 *__restrict__ prim, u64 total_tiles, int op) {
     u64 base = ((u64)blockIdx.x * 4 + (threadIdx.x >> 6)) * 4;
     int lane = threadIdx.x & 63;
     if (base >= total_tiles) return;
@@ -2330,7 +2342,8 @@ __global__ __launch_bounds__(UA_BLOCK) void k_compact(
         u64 room = cap - goff;
         if ((u64)cnt > room) cnt = (u32)room;
         u64 *dst = d.out + goff;
-        const u64 *src = staging + t * stage_stride;
+        const u64 *src = (prim && cnt <= (u32)UA_STAGE_P) ? prim + t * UA_STAGE_P
+                                                          : staging + t * stage_stride;
         for (u32 i = lane; i < cnt; i += 64) dst[i] = src[i];
     }
 }
@@ -2814,8 +2827,8 @@ static thread_local hipError_t g_last_hip = hipSuccess;
 
 enum {
     WS_DESC = 0, WS_TB, WS_TPAIR, WS_TA0, WS_TCNT, WS_TOFF, WS_PARTIAL,
-    WS_STAGE, WS_POUT, WS_HU, WS_HV, WS_HOUT, WS_PACK, WS_SCRATCH_A, WS_SCRATCH_B,
-    WS_LBF, WS_COUNT
+    WS_STAGE, WS_STAGEP, WS_POUT, WS_HU, WS_HV, WS_HOUT, WS_PACK, WS_SCRATCH_A,
+    WS_SCRATCH_B, WS_LBF, WS_COUNT
 };
 
 struct ua_ctx {
@@ -3140,6 +3153,7 @@ static int run_batch_locked(ua_ctx *c, const ua_dpair *pairs, int n_pairs,
         else if (op == OP_DIFF) stage_stride = UA_TILE;
         if (stage_stride) {
             if ((rc = ws_reserve(c, WS_STAGE, total_tiles * stage_stride * sizeof(u64)))) return rc;
+            if ((rc = ws_reserve(c, WS_STAGEP, total_tiles * UA_STAGE_P * sizeof(u64)))) return rc;
         }
     }
 
@@ -3213,16 +3227,19 @@ static int run_batch_locked(ua_ctx *c, const ua_dpair *pairs, int n_pairs,
             }
             HIP_TRY(hipEventRecord(c->ev[1], c->stream));
         } else {
+            const u64 *prim = (!pp_enabled() && !rp_enabled() && op != OP_UNION)
+                                  ? (const u64 *)c->ws[WS_STAGEP]
+                                  : nullptr;
             HIP_TRY(hipEventRecord(c->ev[0], c->stream));
             if (op == OP_INTERSECT) {
                 launch_tiles<OP_INTERSECT, MODE_STAGE>(c, d_descs, d_tpair, d_ta0,
                                                        total_tiles, d_stage,
-                                                       stage_stride, d_tcnt, nullptr,
+                                                       stage_stride, d_tcnt, prim,
                                                        nullptr);
             } else if (op == OP_DIFF) {
                 launch_tiles<OP_DIFF, MODE_STAGE>(c, d_descs, d_tpair, d_ta0,
                                                   total_tiles, d_stage, stage_stride,
-                                                  d_tcnt, nullptr, nullptr);
+                                                  d_tcnt, prim, nullptr);
             } else {
                 launch_tiles<OP_UNION, MODE_COUNT>(c, d_descs, d_tpair, d_ta0,
                                                    total_tiles, nullptr, 0, d_tcnt,
@@ -3244,11 +3261,12 @@ static int run_batch_locked(ua_ctx *c, const ua_dpair *pairs, int n_pairs,
                 u64 cblk = (total_tiles + 15) / 16;
                 hipLaunchKernelGGL(k_compact, dim3((u32)cblk), dim3(UA_BLOCK), 0,
                                    c->stream, d_descs, d_tpair, d_tb, d_tcnt, d_toff,
-                                   d_part, d_stage, stage_stride, total_tiles, op);
+                                   d_part, d_stage, stage_stride, prim, total_tiles,
+                                   op);
                 u64 sblk = (total_tiles + UA_BLOCK - 1) / UA_BLOCK;
                 hipLaunchKernelGGL(k_compact_small, dim3((u32)sblk), dim3(UA_BLOCK),
                                    0, c->stream, d_descs, d_tpair, d_tb, d_tcnt,
-                                   d_toff, d_part, d_stage, stage_stride,
+                                   d_toff, d_part, d_stage, stage_stride, prim,
                                    total_tiles, op);
             }
             u64 poutblk = ((u64)n_pairs + UA_BLOCK - 1) / UA_BLOCK;
@@ -3302,6 +3320,7 @@ struct ua_batch {
     u64 nchunks = 0;
     void *mem = nullptr;     /* one allocation for all metadata arrays */
     u64 *d_stage = nullptr;  /* lazy; stride UA_TILE (fits intersect + diff) */
+    u64 *d_stage_p = nullptr; /* lazy dense primary staging (UA_STAGE_P/tile) */
     /* set-1 buffers for the overlapped run_n (aux tail of step i on the
      * tail stream while step i+1's tile kernel fills the other set) */
     u64 *d_stage1 = nullptr;
@@ -3454,6 +3473,7 @@ extern "C" void ua_batch_destroy(ua_ctx *c, ua_batch *b) {
     if (b->mem) (void)hipFree(b->mem);
     if (b->mem1) (void)hipFree(b->mem1);
     if (b->d_stage) (void)hipFree(b->d_stage);
+    if (b->d_stage_p) (void)hipFree(b->d_stage_p);
     if (b->d_stage1) (void)hipFree(b->d_stage1);
     if (b->d_isplit) (void)hipFree(b->d_isplit);
     delete b;
@@ -3528,13 +3548,27 @@ static int batch_tiles_seq(ua_ctx *c, ua_batch *b, int kop, u64 stride,
         return UA_OK;
     }
     if (record_events) HIP_TRY(hipEventRecord(c->ev[0], c->stream));
+    /* dense primary staging rides the (otherwise unused) offs arg of the
+     * STAGE launch; only for the default kernel and set 0 (overlap keeps
+     * the overflow-only layout, and the experimental kernels ignore it) */
+    const u64 *prim = nullptr;
+    if (set == 0 && !pp_enabled() && !rp_enabled() && kop != OP_UNION) {
+        if (!b->d_stage_p) {
+            size_t pbytes = (size_t)(T ? T : 1) * UA_STAGE_P * sizeof(u64);
+            if (hipMalloc((void **)&b->d_stage_p, pbytes) != hipSuccess) {
+                b->d_stage_p = nullptr;
+                (void)hipGetLastError();
+            }
+        }
+        prim = b->d_stage_p;
+    }
     if (kop == OP_INTERSECT) {
         launch_tiles<OP_INTERSECT, MODE_STAGE>(c, b->d_descs, b->d_tpair, b->d_ta0,
                                                T, stage, stride, tcnt,
-                                               nullptr, nullptr, isin, isout);
+                                               prim, nullptr, isin, isout);
     } else if (kop == OP_DIFF) {
         launch_tiles<OP_DIFF, MODE_STAGE>(c, b->d_descs, b->d_tpair, b->d_ta0, T,
-                                          stage, stride, tcnt, nullptr,
+                                          stage, stride, tcnt, prim,
                                           nullptr, isin, isout);
     } else {
         launch_tiles<OP_UNION, MODE_COUNT>(c, b->d_descs, b->d_tpair, b->d_ta0, T,
@@ -3576,12 +3610,15 @@ static int batch_tail_seq(ua_ctx *c, ua_batch *b, int kop, u64 stride,
                                            b->d_part, isw, nullptr);
         if (record_events) HIP_TRY(hipEventRecord(c->ev[3], st));
     } else {
+        const u64 *prim = (set == 0 && !pp_enabled() && !rp_enabled())
+                              ? b->d_stage_p
+                              : nullptr;
         hipLaunchKernelGGL(k_compact, dim3((u32)((T + 15) / 16)), dim3(UA_BLOCK), 0,
                            st, b->d_descs, tpair, tbx, tcnt, toff,
-                           part, stage, stride_x, T, kop);
+                           part, stage, stride_x, prim, T, kop);
         hipLaunchKernelGGL(k_compact_small, dim3((u32)((T + UA_BLOCK - 1) / UA_BLOCK)),
                            dim3(UA_BLOCK), 0, st, b->d_descs, tpair, tbx, tcnt,
-                           toff, part, stage, stride_x, T, kop);
+                           toff, part, stage, stride_x, prim, T, kop);
     }
     u64 poutblk = ((u64)b->n_pairs + UA_BLOCK - 1) / UA_BLOCK;
     hipLaunchKernelGGL(k_pair_out, dim3((u32)poutblk), dim3(UA_BLOCK), 0, st,
